@@ -1,0 +1,115 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: fused-ABFT SGEMM (abft_kernel_huge) at
+N=4096 fp32 on 1..8 MI355X GPUs (weak scaling: one independent GEMM per GPU,
+the BASELINE.json headline config).
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; for N>1 it
+is launched under torch.distributed.run with one rank per GPU (RCCL).  Rank 0
+prints ONE JSON line; `value` is the whole-job aggregate GFLOPS over all
+ranks, timed as the MAX elapsed over ranks, bracketed by barrier +
+torch.cuda.synchronize on both sides.
+
+A "step" is one full fused-ABFT GEMM launch (C = alpha*A*B^T + beta*C at
+M=N=K=4096, alpha=1, beta=-1.5, with the always-on 20-fault injector and
+in-kernel correction — the reference's headline kernel, BASELINE.md
+abft_kernel_huge row).  vs_baseline divides by the reference's published
+4005 GFLOPS/GPU (T4-class hardware) x n_gpus.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+BASELINE_GFLOPS_PER_GPU = 4005.0  # BASELINE.md abft_kernel_huge @ N=4096
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--size", type=int, default=4096)
+    ap.add_argument("--kernel", default="abft_huge",
+                    choices=["abft_huge", "huge", "rocblas"])
+    ap.add_argument("--mode", default="replicated",
+                    choices=["replicated", "blockrow"])
+    args = ap.parse_args()
+
+    if not torch.cuda.is_available():
+        raise SystemExit("bench.py needs a ROCm GPU (run under gpurun)")
+
+    from ft_sgemm_amd import ops
+    from ft_sgemm_amd.parallel import init_from_env
+
+    rank, world = init_from_env()
+    import torch.distributed as dist
+    dev = torch.device("cuda", torch.cuda.current_device())
+    n = args.size
+    torch.manual_seed(10 + rank)
+    a = (torch.rand((n, n), device=dev) * 1.8 - 0.9).contiguous()
+    b = (torch.rand((n, n), device=dev) * 1.8 - 0.9).contiguous()
+    c = torch.zeros((n, n), device=dev)
+
+    if args.kernel == "abft_huge":
+        step = lambda: ops.ft_sgemm("huge", a, b, c, 1.0, -1.5, inject=True)
+    elif args.kernel == "huge":
+        step = lambda: ops.sgemm("huge", a, b, c, 1.0, -1.5)
+    else:
+        step = lambda: ops.rocblas_sgemm(a, b, c, 1.0, -1.5)
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+        torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        dist.barrier()
+        t = torch.tensor([elapsed], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        torch.cuda.synchronize()
+        elapsed = float(t.item())
+
+    flop_per_step = 2.0 * n * n * n
+    agg_gflops = world * flop_per_step * args.steps / elapsed / 1e9
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "fused_abft_sgemm_gflops",
+            "value": round(agg_gflops, 1),
+            "unit": "GFLOPS",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(
+                agg_gflops / (BASELINE_GFLOPS_PER_GPU * world), 3),
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "abft_kernel_huge",
+                "M": n, "N": n, "K": n,
+                "alpha": 1.0, "beta": -1.5,
+                "inject": True, "faults_per_gemm": 20,
+                "parallelism": f"dp{world}",
+            },
+        }))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

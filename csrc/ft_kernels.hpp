@@ -1,0 +1,283 @@
+// ft_kernels.hpp — MI355X (gfx950/CDNA4) fault-tolerant SGEMM kernel family.
+//
+// Brand-new design with the capabilities of the reference CUDA kernel library
+// (/root/reference/kernel/ft_sgemm/include_code_gen/*.cuh, algorithm traced
+// in SURVEY.md §2.3), re-architected for CDNA4:
+//
+//   * f32-input MFMA outer products (v_mfma_f32_32x32x2_f32 for the
+//     32x32-fragment tiers, v_mfma_f32_16x16x4_f32 for the small tier) —
+//     exact fp32 numerics at the 157 TF/s f32 vector rate.
+//   * LDS double-buffered A/B K-panels staged with global_load_lds
+//     (async HBM->LDS DMA, 16 B per lane), one __syncthreads per K-panel.
+//   * 64-lane-wavefront ABFT: the row/column checksums of each wave's
+//     output sub-tile are maintained per-lane in registers and reduced with
+//     cross-lane butterflies (ds_swizzle/ds_bpermute) — the reference's
+//     block-wide LDS transpose-reduce (ft_sgemm_huge.cuh:346-414) is not
+//     needed because the MFMA accumulator layout makes every column residual
+//     naturally lane-local (acc column == lane % MM) and row residuals
+//     reachable with one bpermute per accumulator register.
+//   * periodic in-kernel verify -> locate (row x col residual intersection)
+//     -> branch-free in-register correction, and a deterministic rotating
+//     fault injector (template flag, not hard-coded: SURVEY.md §5 asks for
+//     measurable overhead with and without injection).
+//
+// Matrix semantics (reference parity, sgemm.cu:108): C = alpha*A*B^T + beta*C,
+// A MxK / B NxK / C MxN, all column-major.  Host launchers require
+// M % BM == 0, N % BN == 0, K % BK == 0 and M,N multiples of 4 (the sweep
+// sizes 1024..6144 step 512 and the distributed N=32768 all satisfy this).
+
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+namespace ftsgemm {
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+template <int MM> struct mfma_traits;
+
+// v_mfma_f32_32x32x2_f32: lane l holds A[i=l&31][k=l>>5], B[k=l>>5][j=l&31];
+// C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5), reg in [0,16).
+template <> struct mfma_traits<32> {
+  static constexpr int kstep = 2;
+  static constexpr int nreg = 16;
+  using acc_t = f32x16;
+  static __device__ inline acc_t mma(float a, float b, acc_t c) {
+    return __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, c, 0, 0, 0);
+  }
+};
+
+// v_mfma_f32_16x16x4_f32: lane l holds A[i=l&15][k=l>>4], B[k=l>>4][j=l&15];
+// C/D: col = lane&15, row = (lane>>4)*4 + reg, reg in [0,4).
+template <> struct mfma_traits<16> {
+  static constexpr int kstep = 4;
+  static constexpr int nreg = 4;
+  using acc_t = f32x4;
+  static __device__ inline acc_t mma(float a, float b, acc_t c) {
+    return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+};
+
+// Butterfly sum across the MM-lane sub-groups (sums a value over the lanes
+// that share one k-slice): masks 1..MM/2 stay inside the group.
+template <int MM> __device__ inline float group_sum(float v) {
+#pragma unroll
+  for (int m = 1; m < MM; m <<= 1) v += __shfl_xor(v, m, 64);
+  return v;
+}
+
+// Completes a per-k-slice partial into a full sum across the 64/MM slices
+// (masks MM..32).
+template <int MM> __device__ inline float slice_sum(float v) {
+#pragma unroll
+  for (int m = MM; m < 64; m <<= 1) v += __shfl_xor(v, m, 64);
+  return v;
+}
+
+// Accumulator register -> row within the MM x MM fragment (valid for both
+// MFMA shapes used here; for MM=16, reg>>2 == 0).
+__device__ constexpr int acc_row(int reg, int sub) {
+  return (reg & 3) + 8 * (reg >> 2) + 4 * sub;
+}
+
+template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
+          bool INJECT>
+__global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
+    int M, int N, int K, const float* __restrict__ A,
+    const float* __restrict__ B, float* __restrict__ C, float alpha,
+    float beta, int verify_iters, int inject_stride, float tau,
+    float inj_mag) {
+  using T = mfma_traits<MM>;
+  constexpr int KSTEP = T::kstep;
+  constexpr int NREG = T::nreg;
+  constexpr int WAVES_M = BM / WM, WAVES_N = BN / WN;
+  constexpr int NWAVES = WAVES_M * WAVES_N;
+  constexpr int THREADS = NWAVES * 64;
+  constexpr int FM = WM / MM, FN = WN / MM;
+  constexpr int BUF = (BM + BN) * BK;  // floats per double-buffer half
+
+  __shared__ float lds[2 * BUF];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int sub = lane / MM;  // k-slice within one MFMA k-step
+  const int r = lane % MM;    // row index of A fragment / col index of B
+  const int wi0 = (wave / WAVES_N) * WM;
+  const int wj0 = (wave % WAVES_N) * WN;
+  const int im0 = blockIdx.x * BM;
+  const int jn0 = blockIdx.y * BN;
+
+  typename T::acc_t acc[FM][FN] = {};
+  float cr[FM] = {};  // running row checksum of this wave's tile (per lane:
+                      // row r of frag fm, k-slices == this lane's sub only)
+  float cc[FN] = {};  // running column checksum (col r of frag fn)
+
+  // ---- async HBM -> LDS staging (global_load_lds, 16 B per lane) ----
+  constexpr int GA = (BM * BK) / (THREADS * 4);  // dwordx4 chunks for A
+  constexpr int GB = (BN * BK) / (THREADS * 4);
+  static_assert(GA >= 1 && GB >= 1, "tile too small for this thread count");
+
+  auto stage = [&](int q, int k0) {
+    float* dstA = &lds[q * BUF];
+    float* dstB = &lds[q * BUF + BM * BK];
+#pragma unroll
+    for (int t = 0; t < GA; ++t) {
+      const int f = (t * THREADS + tid) * 4;  // per-lane float index
+      const int k = f / BM, i = f % BM;       // LDS layout [k][i], i contig
+      const float* g = A + (im0 + i) + (size_t)(k0 + k) * M;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)g,
+          (__attribute__((address_space(3))) void*)(dstA +
+                                                    (t * THREADS + wave * 64) *
+                                                        4),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int t = 0; t < GB; ++t) {
+      const int f = (t * THREADS + tid) * 4;
+      const int k = f / BN, j = f % BN;
+      const float* g = B + (jn0 + j) + (size_t)(k0 + k) * N;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)g,
+          (__attribute__((address_space(3))) void*)(dstB +
+                                                    (t * THREADS + wave * 64) *
+                                                        4),
+          16, 0, 0);
+    }
+  };
+
+  // ---- ABFT verify / locate / correct: wave-autonomous, registers only ----
+  auto verify_correct = [&]() {
+    // Column residuals: acc column == lane's r, so rc is lane-local.
+    float rc[FN];
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      float colp = 0.f;
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int reg = 0; reg < NREG; ++reg) colp += acc[fm][fn][reg];
+      rc[fn] = slice_sum<MM>(colp) - slice_sum<MM>(cc[fn]);
+    }
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+      // Row sums of this wave's frag-row fm, reduced over all wave columns.
+      float rowp[NREG];
+#pragma unroll
+      for (int reg = 0; reg < NREG; ++reg) {
+        float v = acc[fm][0][reg];
+#pragma unroll
+        for (int fn = 1; fn < FN; ++fn) v += acc[fm][fn][reg];
+        rowp[reg] = group_sum<MM>(v);
+      }
+      const float crf = slice_sum<MM>(cr[fm]);  // full checksum for row r
+      // Row residual per accumulator register: fetch the checksum of the
+      // row this register holds from the lane that owns it (lane index ==
+      // row index in sub-group 0).
+      float rr[NREG];
+#pragma unroll
+      for (int reg = 0; reg < NREG; ++reg)
+        rr[reg] = rowp[reg] - __shfl(crf, acc_row(reg, sub), 64);
+      // Branch-free correction at row x column residual intersections
+      // (reference: ft_sgemm_huge.cuh:422-485; sign: residual = computed -
+      // checksum = +error, so subtract).
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn) {
+        const bool cbad = fabsf(rc[fn]) > tau;
+#pragma unroll
+        for (int reg = 0; reg < NREG; ++reg) {
+          const bool bad = cbad && (fabsf(rr[reg]) > tau);
+          acc[fm][fn][reg] -= bad ? rr[reg] : 0.f;
+        }
+      }
+    }
+  };
+
+  // ---- main K loop: one barrier per BK panel, glds prefetch overlaps ----
+  stage(0, 0);
+  __syncthreads();  // drains the in-flight glds (vmcnt(0) inside)
+
+  const int niter = K / BK;
+  for (int it = 0; it < niter; ++it) {
+    const int q = it & 1;
+    if (it + 1 < niter) stage(q ^ 1, (it + 1) * BK);
+    const float* As = &lds[q * BUF];
+    const float* Bs = &lds[q * BUF + BM * BK];
+#pragma unroll
+    for (int kk = 0; kk < BK / KSTEP; ++kk) {
+      const int kloc = kk * KSTEP + sub;
+      float a[FM], b[FN];
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+        a[fm] = As[kloc * BM + wi0 + fm * MM + r];
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
+
+      if constexpr (ABFT) {
+        // Encode: one butterfly each for the A-panel column sum and the
+        // B-panel row sum of this k-slice (summed over the wave's frags
+        // first so the shuffle cost is frag-count-independent), then a
+        // single fma per frag into the running checksums
+        // (reference encode: ft_sgemm_huge.cuh:150-213, redesigned).
+        float asum = a[0], bsum = b[0];
+#pragma unroll
+        for (int fm = 1; fm < FM; ++fm) asum += a[fm];
+#pragma unroll
+        for (int fn = 1; fn < FN; ++fn) bsum += b[fn];
+        const float sa = group_sum<MM>(asum);
+        const float sb = group_sum<MM>(bsum);
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm) cr[fm] = fmaf(a[fm], sb, cr[fm]);
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn) cc[fn] = fmaf(sa, b[fn], cc[fn]);
+      }
+
+      if constexpr (INJECT) {
+        // Deterministic rotating injector (reference:
+        // ft_sgemm_huge.cuh:324-327 injects in every block with a rotating
+        // thread id; here the victim thread also rotates across waves).
+        if (kk == 0 && (it % inject_stride) == 0 &&
+            tid == ((it / inject_stride) * 67) % THREADS)
+          acc[0][0][0] += inj_mag;
+      }
+
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn)
+          acc[fm][fn] = T::mma(a[fm], b[fn], acc[fm][fn]);
+    }
+    if constexpr (ABFT) {
+      if (((it + 1) % verify_iters) == 0 || it + 1 == niter) verify_correct();
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: alpha/beta merge, float4 along column-major columns ----
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      const int j = jn0 + wj0 + fn * MM + r;
+      float* colbase = C + (size_t)j * M + im0 + wi0 + fm * MM;
+#pragma unroll
+      for (int g = 0; g < NREG / 4; ++g) {
+        float* p = colbase + 4 * sub + 8 * g;
+        f32x4 out;
+        if (beta != 0.f) {
+          const f32x4 prev = *(const f32x4*)p;
+#pragma unroll
+          for (int u = 0; u < 4; ++u)
+            out[u] = alpha * acc[fm][fn][4 * g + u] + beta * prev[u];
+        } else {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) out[u] = alpha * acc[fm][fn][4 * g + u];
+        }
+        *(f32x4*)p = out;
+      }
+    }
+}
+
+}  // namespace ftsgemm

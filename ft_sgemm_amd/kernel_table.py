@@ -1,0 +1,62 @@
+"""Kernel-id table and CDNA4 tiling specification.
+
+Kernel ids are API-compatible with the reference dispatch table
+(/root/reference/kernel/ft_sgemm/sgemm.cu:110-199 and :235-237):
+
+  0        vendor BLAS (rocBLAS here, cuBLAS in the reference)
+  1..6     plain SGEMM tiers  small / medium / large / tall / wide / huge
+  7..9     unused -> fall back to rocBLAS (reference: cuBLAS fallback)
+  10       non-fused ABFT baseline composed from rocBLAS calls
+  11..16   fused-ABFT tiers   small / medium / large / tall / wide / huge
+
+The tiling is re-derived for CDNA4 (gfx950): 64-lane wavefronts and MFMA
+matrix cores, NOT the reference's 32-thread-warp register tiling
+(/root/reference/kernel/ft_sgemm/code_gen/main.py:8-16 is the reference's
+7-tuple table; ours is a different parameterisation because the per-thread
+mr x nr register tile is replaced by per-wave MFMA fragments).
+
+Fields per tier:
+  bm, bn     block (workgroup) output tile
+  bk         K-depth of one LDS panel (double buffered)
+  wm, wn     per-wave output sub-tile (waves = (bm/wm)*(bn/wn))
+  mfma       'f32_32x32x2' or 'f32_16x16x4' (f32-input MFMA shapes on gfx950)
+"""
+
+from collections import OrderedDict
+
+TILING = OrderedDict(
+    small=dict(bm=16, bn=16, bk=32, wm=16, wn=16, mfma="f32_16x16x4"),
+    medium=dict(bm=32, bn=32, bk=32, wm=32, wn=32, mfma="f32_32x32x2"),
+    large=dict(bm=64, bn=64, bk=32, wm=64, wn=64, mfma="f32_32x32x2"),
+    tall=dict(bm=128, bn=32, bk=32, wm=64, wn=32, mfma="f32_32x32x2"),
+    wide=dict(bm=32, bn=128, bk=32, wm=32, wn=64, mfma="f32_32x32x2"),
+    huge=dict(bm=128, bn=128, bk=32, wm=64, wn=64, mfma="f32_32x32x2"),
+)
+
+TIERS = list(TILING.keys())
+
+
+def threads(tier: str) -> int:
+    t = TILING[tier]
+    return 64 * (t["bm"] // t["wm"]) * (t["bn"] // t["wn"])
+
+
+# id -> (name, tier or None, fused_abft)
+KERNEL_TABLE = {0: ("cublas", None, False)}
+for i, tier in enumerate(TIERS):
+    KERNEL_TABLE[1 + i] = (f"kernel_sgemm_{tier}", tier, False)
+KERNEL_TABLE[10] = ("abft_baseline", None, False)
+for i, tier in enumerate(TIERS):
+    KERNEL_TABLE[11 + i] = (f"abft_kernel_{tier}", tier, True)
+
+# Names in the exact order of the reference perf sweep
+# (/root/reference/kernel/ft_sgemm/sgemm.cu:235-237: ids {0,1..6,10,11..16}).
+PERF_SWEEP_IDS = [0, 1, 2, 3, 4, 5, 6, 10, 11, 12, 13, 14, 15, 16]
+KERNEL_NAMES = {k: v[0] for k, v in KERNEL_TABLE.items()}
+
+# Fault-injection / ABFT constants (parity with the generated reference
+# kernels, include_code_gen/ft_sgemm_huge.cuh:49-51: err_bound1=9500,
+# error_inject=1e4, 20 injections per GEMM at period K/20).
+ERR_BOUND = 9500.0
+ERROR_INJECT = 10000.0
+N_INJECT = 20

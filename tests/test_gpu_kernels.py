@@ -1,0 +1,119 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference.
+
+The FT kernels run with the always-on injector, so passing the tolerance
+check proves in-kernel detect+locate+correct end to end (the reference's
+de-facto integration test).
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from ft_sgemm_amd import ops
+from ft_sgemm_amd.kernel_table import TIERS
+
+ABS_TOL = 1e-2
+REL_TOL = 1e-2
+
+
+def _require_native():
+    # Fail loudly if the HIP extension is missing on a GPU box.
+    assert ops.have_extension(), "HIP extension must be built on a GPU box"
+
+
+def check(ref, got):
+    diff = (ref - got).abs()
+    rel = diff / ref.abs().clamp_min(1e-30)
+    bad = (diff > ABS_TOL) & (rel > REL_TOL)
+    assert not bad.any(), (
+        f"{int(bad.sum())} mismatches, max abs diff {diff.max().item():.4e}")
+
+
+@pytest.mark.parametrize("tier", TIERS)
+@pytest.mark.parametrize("shape", [(256, 256, 256), (512, 384, 640)])
+def test_plain_tier_vs_torch(tier, shape):
+    _require_native()
+    m, n, k = shape
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.sgemm(tier, a, b, c, 1.0, 0.0)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+@pytest.mark.parametrize("tier", TIERS)
+def test_ft_tier_selftest_injection(tier):
+    """Always-on injection + correction must still match the clean product."""
+    _require_native()
+    m = n = k = 512
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm(tier, a, b, c, 1.0, 0.0, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+@pytest.mark.parametrize("tier", TIERS)
+def test_ft_tier_no_injection(tier):
+    _require_native()
+    m = n = k = 384 if tier != "huge" and tier != "tall" else 512
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm(tier, a, b, c, 1.0, 0.0, inject=False)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_alpha_beta_epilogue():
+    _require_native()
+    m = n = k = 512
+    a, b, c = ops.make_operands(m, n, k)
+    c.normal_(generator=None)
+    c0 = c.clone()
+    ref = ops.torch_reference(a, b, c0, 1.0, -1.5)
+    ops.ft_sgemm("huge", a, b, c, 1.0, -1.5, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_large_k_no_false_positive():
+    """Deep-K accumulation: fp32 roundoff must stay below the ABFT
+    threshold (no spurious corrections that would corrupt the output)."""
+    _require_native()
+    m, n, k = 256, 256, 6144
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=False)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_rocblas_oracle():
+    _require_native()
+    m, n, k = 512, 256, 384
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.rocblas_sgemm(a, b, c, 1.0, 0.0)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_baseline_ft_verdict_and_result():
+    _require_native()
+    m = n = k = 512
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, -1.5)
+    c.normal_()
+    ref = ops.torch_reference(a, b, c, 1.0, -1.5)
+    _, (res_row, res_col) = ops.baseline_ft(a, b, c, 1.0, -1.5)
+    torch.cuda.synchronize()
+    check(ref, c)
+    # fault-free verdicts are roundoff-sized squared norms
+    assert res_row < 1.0 and res_col < 1.0
+
+
+def test_native_extension_is_loaded():
+    """Guards against silent eager fallback: the .so must be in-tree."""
+    import ft_sgemm_amd._C as ext
+    assert "ft_sgemm_amd" in ext.__file__

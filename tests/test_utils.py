@@ -1,0 +1,58 @@
+import numpy as np
+
+from ft_sgemm_amd.utils import PerfTable, generate_random_matrix, gflops, verify_matrix
+from ft_sgemm_amd.kernel_table import (KERNEL_NAMES, KERNEL_TABLE,
+                                       PERF_SWEEP_IDS, TILING, threads)
+
+
+def test_generate_random_matrix_range_and_determinism():
+    m1 = generate_random_matrix(64, 32, seed=10)
+    m2 = generate_random_matrix(64, 32, seed=10)
+    assert m1.dtype == np.float32 and m1.shape == (64, 32)
+    assert np.array_equal(m1, m2)
+    assert m1.max() < 0.9 and m1.min() > -0.9
+    assert m1.flags.f_contiguous  # column-major
+
+
+def test_verify_matrix_reference_semantics():
+    ref = np.array([[100.0, 0.001]], dtype=np.float32)
+    # abs diff 0.5 but rel diff 0.005 < 1e-2 -> passes (AND semantics)
+    got = np.array([[100.5, 0.001]], dtype=np.float32)
+    ok, _, _ = verify_matrix(ref, got)
+    assert ok
+    # abs diff 0.005 < 1e-2, rel enormous -> still passes
+    got2 = np.array([[100.0, 0.006]], dtype=np.float32)
+    ok2, _, _ = verify_matrix(ref, got2)
+    assert ok2
+    # both exceeded -> fails
+    got3 = np.array([[103.0, 0.001]], dtype=np.float32)
+    ok3, idx, _ = verify_matrix(ref, got3)
+    assert not ok3 and idx == (0, 0)
+
+
+def test_gflops_protocol():
+    # 2*M*N*K*reps / t  (sgemm.cu:431-435)
+    assert abs(gflops(1000, 1000, 1000, 5, 1.0) - 10.0) < 1e-9
+
+
+def test_kernel_table_parity():
+    assert KERNEL_NAMES[0] == "cublas"
+    assert KERNEL_NAMES[6] == "kernel_sgemm_huge"
+    assert KERNEL_NAMES[10] == "abft_baseline"
+    assert KERNEL_NAMES[16] == "abft_kernel_huge"
+    assert PERF_SWEEP_IDS == [0, 1, 2, 3, 4, 5, 6, 10, 11, 12, 13, 14, 15, 16]
+    assert 7 not in KERNEL_TABLE and 9 not in KERNEL_TABLE
+    # wavefront-64 blocks, reference block-size parity
+    # (64/64/64/128/128/256 threads)
+    assert [threads(t) for t in TILING] == [64, 64, 64, 128, 128, 256]
+    for t in TILING.values():
+        assert t["bm"] % t["wm"] == 0 and t["bn"] % t["wn"] == 0
+
+
+def test_perf_table_format():
+    tb = PerfTable([1024, 1536])
+    tb.add("cublas", {1024: 4695.4, 1536: 5357.1})
+    text = tb.render()
+    lines = text.splitlines()
+    assert lines[0].startswith("Matrix Size|")
+    assert "4695" in lines[1] and lines[1].endswith("|")
