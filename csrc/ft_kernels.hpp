@@ -96,15 +96,22 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
   constexpr int THREADS = NWAVES * 64;
   constexpr int FM = WM / MM, FN = WN / MM;
   constexpr int BUF = (BM + BN) * BK;  // floats per double-buffer half
+  // ABFT panel-sum scratch: per wave-row-range A-column sums and per
+  // wave-col-range B-column sums of the current K panel (single buffered —
+  // produced and consumed between the same two barriers).
+  constexpr int SA_OFF = 2 * BUF;
+  constexpr int SB_OFF = SA_OFF + WAVES_M * BK;
+  constexpr int LDS_FLOATS = ABFT ? (SB_OFF + WAVES_N * BK) : (2 * BUF);
 
-  __shared__ float lds[2 * BUF];
+  __shared__ __attribute__((aligned(16))) float lds[LDS_FLOATS];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int sub = lane / MM;  // k-slice within one MFMA k-step
   const int r = lane % MM;    // row index of A fragment / col index of B
-  const int wi0 = (wave / WAVES_N) * WM;
-  const int wj0 = (wave % WAVES_N) * WN;
+  const int wm_idx = wave / WAVES_N, wn_idx = wave % WAVES_N;
+  const int wi0 = wm_idx * WM;
+  const int wj0 = wn_idx * WN;
   const int im0 = blockIdx.x * BM;
   const int jn0 = blockIdx.y * BN;
 
@@ -144,6 +151,52 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
                                                     (t * THREADS + wave * 64) *
                                                         4),
           16, 0, 0);
+    }
+  };
+
+  // ---- ABFT panel-sum pass: once per K panel, block-cooperative ----
+  // Computes, into LDS, the per-wave-row-range column sums of the A panel
+  // (sa[wmi][k] = sum_i A[wmi-range i][k]) and the per-wave-col-range sums
+  // of the B panel.  This replaces the reference's per-k-step in-loop
+  // shuffle reduction (ft_sgemm_huge.cuh:155-168): one cooperative sweep
+  // of the staged panel per BK iterations instead of 10 serialized
+  // cross-lane ops per MFMA k-step.
+  auto panel_sums = [&](int q) {
+    const float* As = &lds[q * BUF];
+    const float* Bs = &lds[q * BUF + BM * BK];
+    {
+      constexpr int TASKS = WAVES_M * BK;
+      constexpr int TPT = THREADS / TASKS;      // threads per task (pow2)
+      constexpr int CHUNK = WM / TPT;           // floats per thread (4|CHUNK)
+      const int task = tid / TPT, st = tid % TPT;
+      const int wmi = task / BK, k = task % BK;
+      const float* src = As + k * BM + wmi * WM + st * CHUNK;
+      float s = 0.f;
+#pragma unroll
+      for (int u = 0; u < CHUNK / 4; ++u) {
+        const f32x4 v = *(const f32x4*)(src + 4 * u);
+        s += (v[0] + v[1]) + (v[2] + v[3]);
+      }
+#pragma unroll
+      for (int m = 1; m < TPT; m <<= 1) s += __shfl_xor(s, m, 64);
+      if (st == 0) lds[SA_OFF + task] = s;
+    }
+    {
+      constexpr int TASKS = WAVES_N * BK;
+      constexpr int TPT = THREADS / TASKS;
+      constexpr int CHUNK = WN / TPT;
+      const int task = tid / TPT, st = tid % TPT;
+      const int wni = task / BK, k = task % BK;
+      const float* src = Bs + k * BN + wni * WN + st * CHUNK;
+      float s = 0.f;
+#pragma unroll
+      for (int u = 0; u < CHUNK / 4; ++u) {
+        const f32x4 v = *(const f32x4*)(src + 4 * u);
+        s += (v[0] + v[1]) + (v[2] + v[3]);
+      }
+#pragma unroll
+      for (int m = 1; m < TPT; m <<= 1) s += __shfl_xor(s, m, 64);
+      if (st == 0) lds[SB_OFF + task] = s;
     }
   };
 
@@ -201,6 +254,13 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
   const int niter = K / BK;
   for (int it = 0; it < niter; ++it) {
     const int q = it & 1;
+    if constexpr (ABFT) {
+      // Panel sums must be published before the k-loop consumes them; at
+      // this point no glds is in flight (drained at the previous barrier),
+      // so this extra barrier carries no vmcnt drain.
+      panel_sums(q);
+      __syncthreads();
+    }
     if (it + 1 < niter) stage(q ^ 1, (it + 1) * BK);
     const float* As = &lds[q * BUF];
     const float* Bs = &lds[q * BUF + BM * BK];
@@ -216,18 +276,12 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
         b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
 
       if constexpr (ABFT) {
-        // Encode: one butterfly each for the A-panel column sum and the
-        // B-panel row sum of this k-slice (summed over the wave's frags
-        // first so the shuffle cost is frag-count-independent), then a
-        // single fma per frag into the running checksums
-        // (reference encode: ft_sgemm_huge.cuh:150-213, redesigned).
-        float asum = a[0], bsum = b[0];
-#pragma unroll
-        for (int fm = 1; fm < FM; ++fm) asum += a[fm];
-#pragma unroll
-        for (int fn = 1; fn < FN; ++fn) bsum += b[fn];
-        const float sa = group_sum<MM>(asum);
-        const float sb = group_sum<MM>(bsum);
+        // Encode: two broadcast LDS reads (the precomputed panel sums for
+        // this k-slice) + one fma per fragment into the running checksums
+        // (reference encode: ft_sgemm_huge.cuh:150-213, redesigned around
+        // the once-per-panel cooperative sum pass above).
+        const float sa = lds[SA_OFF + wm_idx * BK + kloc];
+        const float sb = lds[SB_OFF + wn_idx * BK + kloc];
 #pragma unroll
         for (int fm = 0; fm < FM; ++fm) cr[fm] = fmaf(a[fm], sb, cr[fm]);
 #pragma unroll
