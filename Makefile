@@ -10,13 +10,18 @@ gen:
 
 cli: bin/ft_sgemm
 
-bin/ft_sgemm: csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip csrc/ft_kernels.hpp csrc/ft_core.h csrc/generated/tile_params.h
+KERNEL_TUS := $(wildcard csrc/generated/kernel_*.hip)
+
+bin/ft_sgemm: csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip csrc/ft_kernels.hpp csrc/tier_launch.hpp csrc/ft_core.h csrc/generated/tile_params.h $(KERNEL_TUS)
 	mkdir -p bin
 	$(HIPCC) -x hip --offload-arch=$(ARCH) -O3 -std=c++17 -Icsrc \
-	  csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip \
+	  csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip $(KERNEL_TUS) \
 	  -L$(ROCM)/lib -lrocblas -o $@
 
 ext:
+	# ninja does not track header deps for hipcc sources: rebuild .hip TUs
+	# whenever a csrc header is newer than the built extension
+	@if [ -n "$$(find csrc -name '*.hpp' -o -name '*.h' -newer ft_sgemm_amd/_C.cpython-310-x86_64-linux-gnu.so 2>/dev/null)" ]; then touch csrc/*.hip csrc/generated/*.hip; fi
 	PYTORCH_ROCM_ARCH=$(ARCH) python3 setup.py build_ext --inplace
 
 clean:

@@ -53,11 +53,11 @@ bool run_kernel(int kid, int M, int N, int K, const float* dA,
                 const ftsgemm::BaselineWorkspace& ws, bool inject) {
   if (kid >= 1 && kid <= 6)
     return ftsgemm::sgemm_tier_launch(kid - 1, false, false, M, N, K, dA, dB,
-                                      dC, alpha, beta, kTau, kInj,
+                                      dC, alpha, beta, kTau, kInj, 20,
                                       0) == hipSuccess;
   if (kid >= 11 && kid <= 16)
     return ftsgemm::sgemm_tier_launch(kid - 11, true, inject, M, N, K, dA,
-                                      dB, dC, alpha, beta, kTau, kInj,
+                                      dB, dC, alpha, beta, kTau, kInj, 20,
                                       0) == hipSuccess;
   if (kid == 10) {
     float r0 = 0, r1 = 0;

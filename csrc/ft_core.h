@@ -12,7 +12,8 @@ namespace ftsgemm {
 hipError_t sgemm_tier_launch(int tier, bool abft, bool inject, int M, int N,
                              int K, const float* A, const float* B, float* C,
                              float alpha, float beta, float tau,
-                             float inj_mag, hipStream_t stream);
+                             float inj_mag, int verify_windows,
+                             hipStream_t stream);
 
 bool sgemm_tier_supported(int tier, int M, int N, int K);
 

@@ -83,7 +83,7 @@ __device__ constexpr int acc_row(int reg, int sub) {
 
 template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
           bool INJECT>
-__global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
+__global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
     float beta, int verify_iters, int inject_stride, float tau,
@@ -125,7 +125,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
   constexpr int GB = (BN * BK) / (THREADS * 4);
   static_assert(GA >= 1 && GB >= 1, "tile too small for this thread count");
 
-  auto stage = [&](int q, int k0) {
+  auto stage = [&](int q, int k0) __attribute__((always_inline)) {
     float* dstA = &lds[q * BUF];
     float* dstB = &lds[q * BUF + BM * BK];
 #pragma unroll
@@ -161,7 +161,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
   // shuffle reduction (ft_sgemm_huge.cuh:155-168): one cooperative sweep
   // of the staged panel per BK iterations instead of 10 serialized
   // cross-lane ops per MFMA k-step.
-  auto panel_sums = [&](int q) {
+  auto panel_sums = [&](int q) __attribute__((always_inline)) {
     const float* As = &lds[q * BUF];
     const float* Bs = &lds[q * BUF + BM * BK];
     {
@@ -201,7 +201,12 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
   };
 
   // ---- ABFT verify / locate / correct: wave-autonomous, registers only ----
-  auto verify_correct = [&]() {
+  // Two-phase: a cheap detect (compare the total tile sum against the total
+  // checksum — ~70 VALU ops) runs every verify window; the full
+  // locate/correct machinery below is entered only when the residual trips
+  // the threshold (i.e. in the wave that actually absorbed a fault), so the
+  // fault-free common case never pays for location.
+  auto locate_correct = [&]() __attribute__((always_inline)) {
     // Column residuals: acc column == lane's r, so rc is lane-local.
     float rc[FN];
 #pragma unroll
@@ -213,100 +218,122 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN)) void sgemm_mfma(
         for (int reg = 0; reg < NREG; ++reg) colp += acc[fm][fn][reg];
       rc[fn] = slice_sum<MM>(colp) - slice_sum<MM>(cc[fn]);
     }
+    // Row residuals and correction, one accumulator register at a time so
+    // the cold path holds only scalars (an array version here is pushed to
+    // scratch by the register allocator and its spill traffic poisons the
+    // hot loop — measured 2.5x).  Per reg: sum over the wave's columns
+    // (butterfly), fetch the row checksum from the lane that owns it
+    // (lane index == row index in sub-group 0), correct at row x column
+    // residual intersections (reference: ft_sgemm_huge.cuh:422-485; sign:
+    // residual = computed - checksum = +error, so subtract).
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm) {
-      // Row sums of this wave's frag-row fm, reduced over all wave columns.
-      float rowp[NREG];
+      const float crf = slice_sum<MM>(cr[fm]);  // full checksum for row r
 #pragma unroll
       for (int reg = 0; reg < NREG; ++reg) {
         float v = acc[fm][0][reg];
 #pragma unroll
         for (int fn = 1; fn < FN; ++fn) v += acc[fm][fn][reg];
-        rowp[reg] = group_sum<MM>(v);
-      }
-      const float crf = slice_sum<MM>(cr[fm]);  // full checksum for row r
-      // Row residual per accumulator register: fetch the checksum of the
-      // row this register holds from the lane that owns it (lane index ==
-      // row index in sub-group 0).
-      float rr[NREG];
+        v = group_sum<MM>(v);
+        const float rr = v - __shfl(crf, acc_row(reg, sub), 64);
+        const bool rbad = fabsf(rr) > tau;
 #pragma unroll
-      for (int reg = 0; reg < NREG; ++reg)
-        rr[reg] = rowp[reg] - __shfl(crf, acc_row(reg, sub), 64);
-      // Branch-free correction at row x column residual intersections
-      // (reference: ft_sgemm_huge.cuh:422-485; sign: residual = computed -
-      // checksum = +error, so subtract).
-#pragma unroll
-      for (int fn = 0; fn < FN; ++fn) {
-        const bool cbad = fabsf(rc[fn]) > tau;
-#pragma unroll
-        for (int reg = 0; reg < NREG; ++reg) {
-          const bool bad = cbad && (fabsf(rr[reg]) > tau);
-          acc[fm][fn][reg] -= bad ? rr[reg] : 0.f;
+        for (int fn = 0; fn < FN; ++fn) {
+          const bool bad = rbad && (fabsf(rc[fn]) > tau);
+          acc[fm][fn][reg] -= bad ? rr : 0.f;
         }
       }
     }
   };
 
+  auto verify_correct = [&]() __attribute__((always_inline)) {
+    // Cheap detect: total tile sum vs total checksum (~70 VALU + 6
+    // shuffles).  The residual is identical across the wave's lanes, so
+    // the branch is uniform; only a wave that actually absorbed a fault
+    // enters the full locate/correct path above.
+    float tot = 0.f, chk = 0.f;
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+      chk += cr[fm];
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+#pragma unroll
+        for (int reg = 0; reg < NREG; ++reg) tot += acc[fm][fn][reg];
+    }
+    float res = tot - chk;
+#pragma unroll
+    for (int m = 1; m < 64; m <<= 1) res += __shfl_xor(res, m, 64);
+    if (__builtin_expect(fabsf(res) > tau, 0)) locate_correct();
+  };
+
   // ---- main K loop: one barrier per BK panel, glds prefetch overlaps ----
+  // Structure: bursts of `verify_iters` panels.  The inner panel loop never
+  // touches the accumulator with VALU code, so the compiler keeps it in
+  // AGPRs with zero v_accvgpr traffic; injection and verify/correct run at
+  // burst boundaries only (a conditional VALU read of acc inside the panel
+  // loop makes hipcc shuttle all 64 accumulator registers AGPR<->VGPR every
+  // panel — measured 2x wall time on the huge tier).
   stage(0, 0);
   __syncthreads();  // drains the in-flight glds (vmcnt(0) inside)
 
   const int niter = K / BK;
-  for (int it = 0; it < niter; ++it) {
-    const int q = it & 1;
-    if constexpr (ABFT) {
-      // Panel sums must be published before the k-loop consumes them; at
-      // this point no glds is in flight (drained at the previous barrier),
-      // so this extra barrier carries no vmcnt drain.
-      panel_sums(q);
-      __syncthreads();
+  int it = 0;
+  while (it < niter) {
+    if constexpr (INJECT) {
+      // Deterministic rotating injector, once per verify window (reference:
+      // ft_sgemm_huge.cuh:324-327 injects in every block with a rotating
+      // thread id; here the victim also rotates across waves).
+      if (tid == ((unsigned)(it / inject_stride) * 67u) % THREADS)
+        acc[0][0][0] += inj_mag;
     }
-    if (it + 1 < niter) stage(q ^ 1, (it + 1) * BK);
-    const float* As = &lds[q * BUF];
-    const float* Bs = &lds[q * BUF + BM * BK];
-#pragma unroll
-    for (int kk = 0; kk < BK / KSTEP; ++kk) {
-      const int kloc = kk * KSTEP + sub;
-      float a[FM], b[FN];
-#pragma unroll
-      for (int fm = 0; fm < FM; ++fm)
-        a[fm] = As[kloc * BM + wi0 + fm * MM + r];
-#pragma unroll
-      for (int fn = 0; fn < FN; ++fn)
-        b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
-
+    const int burst_end = (it + verify_iters < niter) ? it + verify_iters
+                                                      : niter;
+    for (; it < burst_end; ++it) {
+      const int q = it & 1;
       if constexpr (ABFT) {
-        // Encode: two broadcast LDS reads (the precomputed panel sums for
-        // this k-slice) + one fma per fragment into the running checksums
-        // (reference encode: ft_sgemm_huge.cuh:150-213, redesigned around
-        // the once-per-panel cooperative sum pass above).
-        const float sa = lds[SA_OFF + wm_idx * BK + kloc];
-        const float sb = lds[SB_OFF + wn_idx * BK + kloc];
-#pragma unroll
-        for (int fm = 0; fm < FM; ++fm) cr[fm] = fmaf(a[fm], sb, cr[fm]);
-#pragma unroll
-        for (int fn = 0; fn < FN; ++fn) cc[fn] = fmaf(sa, b[fn], cc[fn]);
+        // Panel sums must be published before the k-loop consumes them; at
+        // this point no glds is in flight (drained at the previous
+        // barrier), so this extra barrier carries no vmcnt drain.
+        panel_sums(q);
+        __syncthreads();
       }
-
-      if constexpr (INJECT) {
-        // Deterministic rotating injector (reference:
-        // ft_sgemm_huge.cuh:324-327 injects in every block with a rotating
-        // thread id; here the victim thread also rotates across waves).
-        if (kk == 0 && (it % inject_stride) == 0 &&
-            tid == ((it / inject_stride) * 67) % THREADS)
-          acc[0][0][0] += inj_mag;
-      }
-
+      if (it + 1 < niter) stage(q ^ 1, (it + 1) * BK);
+      const float* As = &lds[q * BUF];
+      const float* Bs = &lds[q * BUF + BM * BK];
 #pragma unroll
-      for (int fm = 0; fm < FM; ++fm)
+      for (int kk = 0; kk < BK / KSTEP; ++kk) {
+        const int kloc = kk * KSTEP + sub;
+        float a[FM], b[FN];
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm)
+          a[fm] = As[kloc * BM + wi0 + fm * MM + r];
 #pragma unroll
         for (int fn = 0; fn < FN; ++fn)
-          acc[fm][fn] = T::mma(a[fm], b[fn], acc[fm][fn]);
+          b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
+
+        if constexpr (ABFT) {
+          // Encode: two broadcast LDS reads (the precomputed panel sums
+          // for this k-slice) + one fma per fragment into the running
+          // checksums (reference encode: ft_sgemm_huge.cuh:150-213,
+          // redesigned around the once-per-panel cooperative sum pass).
+          const float sa = lds[SA_OFF + wm_idx * BK + kloc];
+          const float sb = lds[SB_OFF + wn_idx * BK + kloc];
+#pragma unroll
+          for (int fm = 0; fm < FM; ++fm) cr[fm] = fmaf(a[fm], sb, cr[fm]);
+#pragma unroll
+          for (int fn = 0; fn < FN; ++fn) cc[fn] = fmaf(sa, b[fn], cc[fn]);
+        }
+
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < FN; ++fn)
+            acc[fm][fn] = T::mma(a[fm], b[fn], acc[fm][fn]);
+      }
+      __syncthreads();
     }
-    if constexpr (ABFT) {
-      if (((it + 1) % verify_iters) == 0 || it + 1 == niter) verify_correct();
-    }
-    __syncthreads();
+    if constexpr (ABFT) verify_correct();
+    if constexpr (!ABFT && !INJECT) { /* plain: single burst, no verify */ }
   }
 
   // ---- epilogue: alpha/beta merge, float4 along column-major columns ----

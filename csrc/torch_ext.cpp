@@ -32,7 +32,7 @@ void check_inputs(const at::Tensor& a, const at::Tensor& b,
 
 void sgemm(int64_t tier, bool abft, bool inject, at::Tensor a, at::Tensor b,
            at::Tensor c, double alpha, double beta, double tau,
-           double inj_mag) {
+           double inj_mag, int64_t verify_windows) {
   int64_t M, N, K;
   check_inputs(a, b, c, M, N, K);
   TORCH_CHECK(ftsgemm::sgemm_tier_supported((int)tier, M, N, K),
@@ -43,7 +43,7 @@ void sgemm(int64_t tier, bool abft, bool inject, at::Tensor a, at::Tensor b,
       (int)tier, abft, inject, (int)M, (int)N, (int)K,
       a.const_data_ptr<float>(), b.const_data_ptr<float>(),
       c.mutable_data_ptr<float>(), (float)alpha, (float)beta, (float)tau,
-      (float)inj_mag, stream.stream());
+      (float)inj_mag, (int)verify_windows, stream.stream());
   TORCH_CHECK(err == hipSuccess,
               "ft_sgemm launch failed: ", hipGetErrorString(err));
 }

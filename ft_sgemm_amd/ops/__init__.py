@@ -43,19 +43,19 @@ def sgemm(tier: str, a: torch.Tensor, b: torch.Tensor, c: torch.Tensor,
     """Plain hand-tiled MFMA SGEMM.  a:(K,M) b:(K,N) c:(N,M) fp32 CUDA
     tensors holding column-major A (MxK), B (NxK), C (MxN).  In-place on c."""
     _require_ext().sgemm(tier_index(tier), False, False, a, b, c,
-                         alpha, beta, 0.0, 0.0)
+                         alpha, beta, 0.0, 0.0, 20)
     return c
 
 
 def ft_sgemm(tier: str, a: torch.Tensor, b: torch.Tensor, c: torch.Tensor,
              alpha: float = 1.0, beta: float = 0.0, inject: bool = True,
-             tau: float = ERR_BOUND,
-             inj_mag: float = ERROR_INJECT) -> torch.Tensor:
+             tau: float = ERR_BOUND, inj_mag: float = ERROR_INJECT,
+             verify_windows: int = 20) -> torch.Tensor:
     """Fused-ABFT MFMA SGEMM with in-kernel verify/locate/correct.  The
     default inject=True preserves the reference's always-self-testing
     property."""
     _require_ext().sgemm(tier_index(tier), True, inject, a, b, c,
-                         alpha, beta, tau, inj_mag)
+                         alpha, beta, tau, inj_mag, verify_windows)
     return c
 
 

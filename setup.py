@@ -6,6 +6,7 @@ The built ft_sgemm_amd/_C*.so travels with the repo snapshot to GPU boxes
 (it is git-ignored but not gpurun-ignored).
 """
 
+import glob
 import os
 
 from setuptools import setup
@@ -27,7 +28,7 @@ setup(
                 "csrc/torch_ext.cpp",
                 "csrc/dispatch.hip",
                 "csrc/rocblas_path.hip",
-            ],
+            ] + sorted(glob.glob("csrc/generated/kernel_*.hip")),
             include_dirs=[os.path.join(ROOT, "csrc")],
             libraries=["rocblas"],
             extra_compile_args={

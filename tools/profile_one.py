@@ -20,18 +20,25 @@ def main():
     ap.add_argument("--size", type=int, default=4096)
     ap.add_argument("--reps", type=int, default=10)
     ap.add_argument("--no-inject", action="store_true")
+    ap.add_argument("--vw", type=int, default=20)
     args = ap.parse_args()
     n = args.size
     a, b, c = ops.make_operands(n, n, n)
+    def run():
+        if args.kid in range(11, 17):
+            ops.ft_sgemm(["small","medium","large","tall","wide","huge"][args.kid-11],
+                         a, b, c, 1.0, -1.5, inject=not args.no_inject,
+                         verify_windows=args.vw)
+        else:
+            ops.run_kernel_id(args.kid, a, b, c, 1.0, -1.5,
+                              inject=not args.no_inject)
     for _ in range(2):
-        ops.run_kernel_id(args.kid, a, b, c, 1.0, -1.5,
-                          inject=not args.no_inject)
+        run()
     torch.cuda.synchronize()
     import time
     t0 = time.perf_counter()
     for _ in range(args.reps):
-        ops.run_kernel_id(args.kid, a, b, c, 1.0, -1.5,
-                          inject=not args.no_inject)
+        run()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     gf = 2 * n**3 * args.reps / dt / 1e9
