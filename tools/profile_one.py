@@ -6,11 +6,12 @@
 """
 
 import argparse
+import os
 import sys
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from ft_sgemm_amd import ops  # noqa: E402
 
 
