@@ -48,17 +48,32 @@ const char* kNames[17] = {
 
 constexpr float kTau = 9500.f, kInj = 10000.f;
 
+// fused-ABFT segment-checksum scratch, grown on demand
+float* abft_ws(size_t floats) {
+  static float* buf = nullptr;
+  static size_t cap = 0;
+  if (floats > cap) {
+    if (buf) (void)hipFree(buf);
+    HIP_CALL(hipMalloc(&buf, floats * sizeof(float)));
+    cap = floats;
+  }
+  return buf;
+}
+
 bool run_kernel(int kid, int M, int N, int K, const float* dA,
                 const float* dB, float* dC, float alpha, float beta,
                 const ftsgemm::BaselineWorkspace& ws, bool inject) {
   if (kid >= 1 && kid <= 6)
     return ftsgemm::sgemm_tier_launch(kid - 1, false, false, M, N, K, dA, dB,
                                       dC, alpha, beta, kTau, kInj, 20,
-                                      0) == hipSuccess;
-  if (kid >= 11 && kid <= 16)
+                                      nullptr, 0) == hipSuccess;
+  if (kid >= 11 && kid <= 16) {
+    float* w = abft_ws(
+        ftsgemm::sgemm_abft_workspace_floats(kid - 11, M, N, K));
     return ftsgemm::sgemm_tier_launch(kid - 11, true, inject, M, N, K, dA,
                                       dB, dC, alpha, beta, kTau, kInj, 20,
-                                      0) == hipSuccess;
+                                      w, 0) == hipSuccess;
+  }
   if (kid == 10) {
     float r0 = 0, r1 = 0;
     return ftsgemm::baseline_ft_sgemm(M, N, K, dA, dB, dC, alpha, beta, ws,

@@ -15,25 +15,40 @@ namespace ftsgemm {
   hipError_t launch_tier_##name(bool abft, bool inject, int M, int N, int K, \
                                 const float* A, const float* B, float* C,    \
                                 float alpha, float beta, float tau,          \
-                                float inj_mag, int verify_windows,           \
-                                hipStream_t stream);
+                                float inj_mag, int verify_windows, float* ws,\
+                                hipStream_t stream);                         \
+  size_t abft_ws_floats_##name(int M, int N, int K);
 FT_TIER_LIST(FT_DECL)
 #undef FT_DECL
 
 hipError_t sgemm_tier_launch(int tier, bool abft, bool inject, int M, int N,
                              int K, const float* A, const float* B, float* C,
                              float alpha, float beta, float tau,
-                             float inj_mag, int verify_windows,
+                             float inj_mag, int verify_windows, float* ws,
                              hipStream_t stream) {
   switch (tier) {
 #define FT_CASE(name, BM, BN, BK, WM, WN, MM)                              \
   case FT_TIER_ID_##name:                                                  \
     return launch_tier_##name(abft, inject, M, N, K, A, B, C, alpha, beta, \
-                              tau, inj_mag, verify_windows, stream);
+                              tau, inj_mag, verify_windows, ws, stream);
     FT_TIER_LIST(FT_CASE)
 #undef FT_CASE
     default:
       return hipErrorInvalidValue;
+  }
+}
+
+// Workspace floats needed by the fused-ABFT path of a tier (segment-sum
+// matrices SA+SB); 0 for plain kernels.
+size_t sgemm_abft_workspace_floats(int tier, int M, int N, int K) {
+  switch (tier) {
+#define FT_WS(name, BM, BN, BK, WM, WN, MM) \
+  case FT_TIER_ID_##name:                   \
+    return abft_ws_floats_##name(M, N, K);
+    FT_TIER_LIST(FT_WS)
+#undef FT_WS
+    default:
+      return 0;
   }
 }
 

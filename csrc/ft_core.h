@@ -9,13 +9,18 @@ namespace ftsgemm {
 // Launch one of the six hand-tiled MFMA SGEMM kernels (tier id per
 // generated/tile_params.h; abft selects the fused-ABFT twin, inject the
 // always-on fault injector).  C = alpha*A*B^T + beta*C, column-major.
+// When abft, `ws` must point to a device scratch buffer of
+// sgemm_abft_workspace_floats(tier, M, N, K) fp32 elements (the launcher
+// fills it with the segment checksums of A and B before the fused kernel).
 hipError_t sgemm_tier_launch(int tier, bool abft, bool inject, int M, int N,
                              int K, const float* A, const float* B, float* C,
                              float alpha, float beta, float tau,
-                             float inj_mag, int verify_windows,
+                             float inj_mag, int verify_windows, float* ws,
                              hipStream_t stream);
 
 bool sgemm_tier_supported(int tier, int M, int N, int K);
+
+size_t sgemm_abft_workspace_floats(int tier, int M, int N, int K);
 
 // rocBLAS paths (kernel id 0 oracle and the id-10 non-fused ABFT baseline,
 // reference: cuBLAS at sgemm.cu:108 / include/baseline_ft_sgemm.cuh).

@@ -9,13 +9,24 @@
 //     exact fp32 numerics at the 157 TF/s f32 vector rate.
 //   * LDS double-buffered A/B K-panels staged with global_load_lds
 //     (async HBM->LDS DMA, 16 B per lane), one __syncthreads per K-panel.
-//   * 64-lane-wavefront ABFT: the row/column checksums of each wave's
-//     output sub-tile are maintained per-lane in registers and reduced with
-//     cross-lane butterflies (ds_swizzle/ds_bpermute) — the reference's
-//     block-wide LDS transpose-reduce (ft_sgemm_huge.cuh:346-414) is not
-//     needed because the MFMA accumulator layout makes every column residual
-//     naturally lane-local (acc column == lane % MM) and row residuals
-//     reachable with one bpermute per accumulator register.
+//   * Checksum ENCODE from precomputed segment sums: a separate
+//     bandwidth-bound kernel (segsum_kernel) computes SA[seg][k] =
+//     sum of A rows in each WM-segment (and SB for B) in one coalesced
+//     pass (~2% of GEMM time at N=4096); the fused kernel streams each
+//     wave's 64-k strip of SA/SB into a private LDS strip with one
+//     4-B-per-lane global_load_lds per panel pair.  This replaces the
+//     reference's per-k-step in-loop shuffle encode
+//     (ft_sgemm_huge.cuh:150-213) and the earlier in-kernel cooperative
+//     panel-sums pass: ablation (tools/probe_ablate.hip) measured the
+//     in-kernel sums pass + its extra barrier at -11.6% of plain GEMM
+//     throughput, vs ~2% for the precompute pass.
+//   * 64-lane-wavefront ABFT verify: the row/column checksums of each
+//     wave's output sub-tile are maintained per-lane in registers and
+//     reduced with cross-lane butterflies — the reference's block-wide LDS
+//     transpose-reduce (ft_sgemm_huge.cuh:346-414) is not needed because
+//     the MFMA accumulator layout makes every column residual naturally
+//     lane-local (acc column == lane % MM) and row residuals reachable
+//     with one shuffle per accumulator register.
 //   * periodic in-kernel verify -> locate (row x col residual intersection)
 //     -> branch-free in-register correction, and a deterministic rotating
 //     fault injector (template flag, not hard-coded: SURVEY.md §5 asks for
@@ -81,13 +92,40 @@ __device__ constexpr int acc_row(int reg, int sub) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * sub;
 }
 
+// Segment-sum precompute: SA[seg * sstr + k] = sum_{i in [seg*SEG, +SEG)}
+// A[i + k*M] for a column-major MxK matrix.  One workgroup per column k,
+// coalesced f32x4 sweep, SEG/4-lane shuffle groups, no LDS, no atomics.
+// sstr must be >= K and a multiple of 64 (the fused kernel streams 64-k
+// strips with a 4-B/lane global_load_lds and must not cross rows).
+template <int SEG>
+__global__ __launch_bounds__(256) void segsum_kernel(
+    int M, int K, int sstr, const float* __restrict__ A,
+    float* __restrict__ SA) {
+  const int k = blockIdx.x;
+  const int tid = threadIdx.x;
+  const float* col = A + (size_t)k * M;
+  constexpr int GL = SEG / 4;  // lanes per segment group (4|8|16)
+  for (int base = 0; base < M; base += 1024) {
+    const int idx = base + tid * 4;
+    float s = 0.f;
+    if (idx < M) {
+      const f32x4 v = *(const f32x4*)(col + idx);
+      s = (v[0] + v[1]) + (v[2] + v[3]);
+    }
+#pragma unroll
+    for (int m = 1; m < GL; m <<= 1) s += __shfl_xor(s, m, 64);
+    if (idx < M && (idx % SEG) == 0)
+      SA[(size_t)(idx / SEG) * sstr + k] = s;
+  }
+}
+
 template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
           bool INJECT>
 __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
-    float beta, int verify_iters, int inject_stride, float tau,
-    float inj_mag) {
+    float beta, int verify_iters, int inject_stride, float tau, float inj_mag,
+    const float* __restrict__ SA, const float* __restrict__ SB, int sstr) {
   using T = mfma_traits<MM>;
   constexpr int KSTEP = T::kstep;
   constexpr int NREG = T::nreg;
@@ -96,12 +134,10 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   constexpr int THREADS = NWAVES * 64;
   constexpr int FM = WM / MM, FN = WN / MM;
   constexpr int BUF = (BM + BN) * BK;  // floats per double-buffer half
-  // ABFT panel-sum scratch: per wave-row-range A-column sums and per
-  // wave-col-range B-column sums of the current K panel (single buffered —
-  // produced and consumed between the same two barriers).
-  constexpr int SA_OFF = 2 * BUF;
-  constexpr int SB_OFF = SA_OFF + WAVES_M * BK;
-  constexpr int LDS_FLOATS = ABFT ? (SB_OFF + WAVES_N * BK) : (2 * BUF);
+  // ABFT checksum strips: per wave, a private [2 pair-buffers][sa(64)|sb(64)]
+  // window of the precomputed segment sums, streamed by global_load_lds.
+  constexpr int STRIP_OFF = 2 * BUF;
+  constexpr int LDS_FLOATS = ABFT ? (STRIP_OFF + NWAVES * 256) : (2 * BUF);
 
   __shared__ __attribute__((aligned(16))) float lds[LDS_FLOATS];
 
@@ -154,50 +190,24 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
     }
   };
 
-  // ---- ABFT panel-sum pass: once per K panel, block-cooperative ----
-  // Computes, into LDS, the per-wave-row-range column sums of the A panel
-  // (sa[wmi][k] = sum_i A[wmi-range i][k]) and the per-wave-col-range sums
-  // of the B panel.  This replaces the reference's per-k-step in-loop
-  // shuffle reduction (ft_sgemm_huge.cuh:155-168): one cooperative sweep
-  // of the staged panel per BK iterations instead of 10 serialized
-  // cross-lane ops per MFMA k-step.
-  auto panel_sums = [&](int q) __attribute__((always_inline)) {
-    const float* As = &lds[q * BUF];
-    const float* Bs = &lds[q * BUF + BM * BK];
-    {
-      constexpr int TASKS = WAVES_M * BK;
-      constexpr int TPT = THREADS / TASKS;      // threads per task (pow2)
-      constexpr int CHUNK = WM / TPT;           // floats per thread (4|CHUNK)
-      const int task = tid / TPT, st = tid % TPT;
-      const int wmi = task / BK, k = task % BK;
-      const float* src = As + k * BM + wmi * WM + st * CHUNK;
-      float s = 0.f;
-#pragma unroll
-      for (int u = 0; u < CHUNK / 4; ++u) {
-        const f32x4 v = *(const f32x4*)(src + 4 * u);
-        s += (v[0] + v[1]) + (v[2] + v[3]);
-      }
-#pragma unroll
-      for (int m = 1; m < TPT; m <<= 1) s += __shfl_xor(s, m, 64);
-      if (st == 0) lds[SA_OFF + task] = s;
-    }
-    {
-      constexpr int TASKS = WAVES_N * BK;
-      constexpr int TPT = THREADS / TASKS;
-      constexpr int CHUNK = WN / TPT;
-      const int task = tid / TPT, st = tid % TPT;
-      const int wni = task / BK, k = task % BK;
-      const float* src = Bs + k * BN + wni * WN + st * CHUNK;
-      float s = 0.f;
-#pragma unroll
-      for (int u = 0; u < CHUNK / 4; ++u) {
-        const f32x4 v = *(const f32x4*)(src + 4 * u);
-        s += (v[0] + v[1]) + (v[2] + v[3]);
-      }
-#pragma unroll
-      for (int m = 1; m < TPT; m <<= 1) s += __shfl_xor(s, m, 64);
-      if (st == 0) lds[SB_OFF + task] = s;
-    }
+  // ---- ABFT strip staging: one 4-B/lane glds pair per TWO K panels ----
+  // Each wave streams its own 64-k window of the precomputed segment sums
+  // (SA row of its WM range, SB row of its WN range) into a private LDS
+  // strip.  No extra barrier, no cooperative pass: the strips ride the
+  // same prefetch pipeline as the A/B panels and are drained by the same
+  // end-of-panel __syncthreads.
+  const int segA = blockIdx.x * WAVES_M + wm_idx;
+  const int segB = blockIdx.y * WAVES_N + wn_idx;
+  auto strip_stage = [&](int pb, int k0) __attribute__((always_inline)) {
+    const float* ga = SA + (size_t)segA * sstr + k0;
+    const float* gb = SB + (size_t)segB * sstr + k0;
+    float* dst = &lds[STRIP_OFF + wave * 256 + pb * 128];
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)ga,
+        (__attribute__((address_space(3))) void*)dst, 4, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gb,
+        (__attribute__((address_space(3))) void*)(dst + 64), 4, 0, 0);
   };
 
   // ---- ABFT verify / locate / correct: wave-autonomous, registers only ----
@@ -274,6 +284,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   // loop makes hipcc shuttle all 64 accumulator registers AGPR<->VGPR every
   // panel — measured 2x wall time on the huge tier).
   stage(0, 0);
+  if constexpr (ABFT) strip_stage(0, 0);  // pair (panels 0,1)
   __syncthreads();  // drains the in-flight glds (vmcnt(0) inside)
 
   const int niter = K / BK;
@@ -290,16 +301,20 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
                                                       : niter;
     for (; it < burst_end; ++it) {
       const int q = it & 1;
-      if constexpr (ABFT) {
-        // Panel sums must be published before the k-loop consumes them; at
-        // this point no glds is in flight (drained at the previous
-        // barrier), so this extra barrier carries no vmcnt drain.
-        panel_sums(q);
-        __syncthreads();
+      if (it + 1 < niter) {
+        stage(q ^ 1, (it + 1) * BK);
+        // a new strip pair rides along every other panel prefetch
+        if constexpr (ABFT) {
+          if (((it + 1) & 1) == 0) strip_stage(((it + 1) >> 1) & 1,
+                                               (it + 1) * BK);
+        }
       }
-      if (it + 1 < niter) stage(q ^ 1, (it + 1) * BK);
       const float* As = &lds[q * BUF];
       const float* Bs = &lds[q * BUF + BM * BK];
+      const float* strip =
+          ABFT ? &lds[STRIP_OFF + wave * 256 + ((it >> 1) & 1) * 128 +
+                      (it & 1) * 32]
+               : nullptr;
 #pragma unroll
       for (int kk = 0; kk < BK / KSTEP; ++kk) {
         const int kloc = kk * KSTEP + sub;
@@ -312,12 +327,12 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
           b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
 
         if constexpr (ABFT) {
-          // Encode: two broadcast LDS reads (the precomputed panel sums
-          // for this k-slice) + one fma per fragment into the running
+          // Encode: two broadcast LDS reads of the precomputed segment
+          // sums for this k-slice + one fma per fragment into the running
           // checksums (reference encode: ft_sgemm_huge.cuh:150-213,
-          // redesigned around the once-per-panel cooperative sum pass).
-          const float sa = lds[SA_OFF + wm_idx * BK + kloc];
-          const float sb = lds[SB_OFF + wn_idx * BK + kloc];
+          // redesigned around the offline segsum pass).
+          const float sa = strip[kloc];
+          const float sb = strip[64 + kloc];
 #pragma unroll
           for (int fm = 0; fm < FM; ++fm) cr[fm] = fmaf(a[fm], sb, cr[fm]);
 #pragma unroll

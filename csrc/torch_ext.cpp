@@ -39,11 +39,20 @@ void sgemm(int64_t tier, bool abft, bool inject, at::Tensor a, at::Tensor b,
               "ft_sgemm: tier ", tier, " requires M,N,K multiples of its "
               "tile (M=", M, " N=", N, " K=", K, ")");
   auto stream = at::cuda::getCurrentCUDAStream();
+  at::Tensor ws;
+  float* ws_ptr = nullptr;
+  if (abft) {
+    // segment-checksum scratch (SA/SB), via the caching allocator
+    size_t n = ftsgemm::sgemm_abft_workspace_floats((int)tier, (int)M,
+                                                    (int)N, (int)K);
+    ws = at::empty({(int64_t)n}, a.options());
+    ws_ptr = ws.mutable_data_ptr<float>();
+  }
   hipError_t err = ftsgemm::sgemm_tier_launch(
       (int)tier, abft, inject, (int)M, (int)N, (int)K,
       a.const_data_ptr<float>(), b.const_data_ptr<float>(),
       c.mutable_data_ptr<float>(), (float)alpha, (float)beta, (float)tau,
-      (float)inj_mag, (int)verify_windows, stream.stream());
+      (float)inj_mag, (int)verify_windows, ws_ptr, stream.stream());
   TORCH_CHECK(err == hipSuccess,
               "ft_sgemm launch failed: ", hipGetErrorString(err));
 }
