@@ -10,27 +10,27 @@
 //   * LDS double-buffered A/B K-panels staged with global_load_lds
 //     (async HBM->LDS DMA, 16 B per lane), one __syncthreads per K-panel.
 //   * Checksum ENCODE from precomputed segment sums: a separate
-//     bandwidth-bound kernel (segsum_kernel) computes SA[seg][k] =
-//     sum of A rows in each WM-segment (and SB for B) in one coalesced
-//     pass (~2% of GEMM time at N=4096); the fused kernel streams each
-//     wave's 64-k strip of SA/SB into a private LDS strip with one
+//     bandwidth-bound kernel (segsum_kernel) computes, per WM-row band of
+//     A, the plain and the row-index-weighted column sums in one coalesced
+//     pass (~1-2% of GEMM time at N=4096); the fused kernel streams each
+//     wave's 64-k strip of them into a private LDS strip with one
 //     4-B-per-lane global_load_lds per panel pair.  This replaces the
 //     reference's per-k-step in-loop shuffle encode
 //     (ft_sgemm_huge.cuh:150-213) and the earlier in-kernel cooperative
 //     panel-sums pass: ablation (tools/probe_ablate.hip) measured the
 //     in-kernel sums pass + its extra barrier at -11.6% of plain GEMM
 //     throughput, vs ~2% for the precompute pass.
-//   * 64-lane-wavefront ABFT verify: the row/column checksums of each
-//     wave's output sub-tile are maintained per-lane in registers and
-//     reduced with cross-lane butterflies — the reference's block-wide LDS
-//     transpose-reduce (ft_sgemm_huge.cuh:346-414) is not needed because
-//     the MFMA accumulator layout makes every column residual naturally
-//     lane-local (acc column == lane % MM) and row residuals reachable
-//     with one shuffle per accumulator register.
-//   * periodic in-kernel verify -> locate (row x col residual intersection)
-//     -> branch-free in-register correction, and a deterministic rotating
-//     fault injector (template flag, not hard-coded: SURVEY.md §5 asks for
-//     measurable overhead with and without injection).
+//   * 64-lane-wavefront ABFT verify with RATIO LOCATE: each wave maintains
+//     per-lane column checksums (plain cc + row-weighted cw) of its output
+//     sub-tile.  A fault's column is lane-local (MFMA acc column == lane %
+//     MM), its magnitude is the plain column residual rc, and its row is
+//     round(rw/rc) — no cross-lane row reductions at all, unlike the
+//     reference's block-wide LDS transpose-reduce + row x col intersection
+//     (ft_sgemm_huge.cuh:346-485).
+//   * periodic in-kernel verify -> ratio locate -> branch-free in-register
+//     correction, and a deterministic rotating fault injector (template
+//     flag, not hard-coded: SURVEY.md §5 asks for measurable overhead with
+//     and without injection).
 //
 // Matrix semantics (reference parity, sgemm.cu:108): C = alpha*A*B^T + beta*C,
 // A MxK / B NxK / C MxN, all column-major.  Host launchers require
@@ -93,29 +93,39 @@ __device__ constexpr int acc_row(int reg, int sub) {
 }
 
 // Segment-sum precompute: SA[seg * sstr + k] = sum_{i in [seg*SEG, +SEG)}
-// A[i + k*M] for a column-major MxK matrix.  One workgroup per column k,
-// coalesced f32x4 sweep, SEG/4-lane shuffle groups, no LDS, no atomics.
-// sstr must be >= K and a multiple of 64 (the fused kernel streams 64-k
-// strips with a 4-B/lane global_load_lds and must not cross rows).
+// A[i + k*M] for a column-major MxK matrix, and (when SW != nullptr) the
+// row-index-weighted sums SW[seg * sstr + k] = sum_i (i % SEG) * A[i + k*M]
+// used by the ratio locate (fault row = round(rw / rc)).  One workgroup per
+// column k, coalesced f32x4 sweep, SEG/4-lane shuffle groups, no LDS, no
+// atomics.  sstr must be >= K and a multiple of 64 (the fused kernel
+// streams 64-k strips with a 4-B/lane global_load_lds, never crossing rows).
 template <int SEG>
 __global__ __launch_bounds__(256) void segsum_kernel(
     int M, int K, int sstr, const float* __restrict__ A,
-    float* __restrict__ SA) {
+    float* __restrict__ SA, float* __restrict__ SW) {
   const int k = blockIdx.x;
   const int tid = threadIdx.x;
   const float* col = A + (size_t)k * M;
   constexpr int GL = SEG / 4;  // lanes per segment group (4|8|16)
   for (int base = 0; base < M; base += 1024) {
     const int idx = base + tid * 4;
-    float s = 0.f;
+    float s = 0.f, w = 0.f;
     if (idx < M) {
       const f32x4 v = *(const f32x4*)(col + idx);
       s = (v[0] + v[1]) + (v[2] + v[3]);
+      const float i0 = (float)(idx % SEG);
+      w = i0 * v[0] + (i0 + 1.f) * v[1] + (i0 + 2.f) * v[2] +
+          (i0 + 3.f) * v[3];
     }
 #pragma unroll
-    for (int m = 1; m < GL; m <<= 1) s += __shfl_xor(s, m, 64);
-    if (idx < M && (idx % SEG) == 0)
+    for (int m = 1; m < GL; m <<= 1) {
+      s += __shfl_xor(s, m, 64);
+      w += __shfl_xor(w, m, 64);
+    }
+    if (idx < M && (idx % SEG) == 0) {
       SA[(size_t)(idx / SEG) * sstr + k] = s;
+      if (SW) SW[(size_t)(idx / SEG) * sstr + k] = w;
+    }
   }
 }
 
@@ -125,7 +135,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
     float beta, int verify_iters, int inject_stride, float tau, float inj_mag,
-    const float* __restrict__ SA, const float* __restrict__ SB, int sstr) {
+    const float* __restrict__ SA, int sstr) {
   using T = mfma_traits<MM>;
   constexpr int KSTEP = T::kstep;
   constexpr int NREG = T::nreg;
@@ -134,8 +144,10 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   constexpr int THREADS = NWAVES * 64;
   constexpr int FM = WM / MM, FN = WN / MM;
   constexpr int BUF = (BM + BN) * BK;  // floats per double-buffer half
-  // ABFT checksum strips: per wave, a private [2 pair-buffers][sa(64)|sb(64)]
-  // window of the precomputed segment sums, streamed by global_load_lds.
+  // ABFT checksum strips: per wave, a private [2 pair-buffers][sa(64) |
+  // saw(64)] window of the precomputed segment sums, streamed by
+  // global_load_lds.  Only A-side sums are needed: the column checksums
+  // detect, the weighted ones locate the row, and the column is lane-local.
   constexpr int STRIP_OFF = 2 * BUF;
   constexpr int LDS_FLOATS = ABFT ? (STRIP_OFF + NWAVES * 256) : (2 * BUF);
 
@@ -152,9 +164,10 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   const int jn0 = blockIdx.y * BN;
 
   typename T::acc_t acc[FM][FN] = {};
-  float cr[FM] = {};  // running row checksum of this wave's tile (per lane:
-                      // row r of frag fm, k-slices == this lane's sub only)
-  float cc[FN] = {};  // running column checksum (col r of frag fn)
+  float cc[FN] = {};  // running column checksum of this wave's tile (per
+                      // lane: col r of frag fn, this lane's k-slices only)
+  float cw[FN] = {};  // row-index-weighted column checksum (ratio locate:
+                      // fault row = round(residual(cw) / residual(cc)))
 
   // ---- async HBM -> LDS staging (global_load_lds, 16 B per lane) ----
   constexpr int GA = (BM * BK) / (THREADS * 4);  // dwordx4 chunks for A
@@ -196,63 +209,62 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   // strip.  No extra barrier, no cooperative pass: the strips ride the
   // same prefetch pipeline as the A/B panels and are drained by the same
   // end-of-panel __syncthreads.
+  // Workspace layout (set up by the launcher): SA = [plain A-segment sums
+  // (M/WM rows) | row-weighted A-segment sums (M/WM rows)]; each row sstr
+  // floats.  SB is unused by this scheme and passed null.
   const int segA = blockIdx.x * WAVES_M + wm_idx;
-  const int segB = blockIdx.y * WAVES_N + wn_idx;
+  const int segsA = ABFT ? (M / WM) : 0;
   auto strip_stage = [&](int pb, int k0) __attribute__((always_inline)) {
-    const float* ga = SA + (size_t)segA * sstr + k0;
-    const float* gb = SB + (size_t)segB * sstr + k0;
+    // glds source addresses are PER-LANE (the LDS side is uniform base +
+    // lane*size): lane l fetches SA[segA][k0 + l] into strip slot l.
+    const float* ga = SA + (size_t)segA * sstr + k0 + lane;
+    const float* gw = SA + (size_t)(segsA + segA) * sstr + k0 + lane;
     float* dst = &lds[STRIP_OFF + wave * 256 + pb * 128];
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)ga,
         (__attribute__((address_space(3))) void*)dst, 4, 0, 0);
     __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)gb,
+        (const __attribute__((address_space(1))) void*)gw,
         (__attribute__((address_space(3))) void*)(dst + 64), 4, 0, 0);
   };
 
   // ---- ABFT verify / locate / correct: wave-autonomous, registers only ----
   // Two-phase: a cheap detect (compare the total tile sum against the total
-  // checksum — ~70 VALU ops) runs every verify window; the full
-  // locate/correct machinery below is entered only when the residual trips
-  // the threshold (i.e. in the wave that actually absorbed a fault), so the
-  // fault-free common case never pays for location.
+  // checksum — ~70 VALU ops) runs every verify window; the locate/correct
+  // below is entered only when the residual trips the threshold (i.e. in
+  // the wave that actually absorbed a fault), so the fault-free common case
+  // never pays for location.
+  //
+  // Ratio locate: the fault's COLUMN is lane-local (acc column == lane's
+  // r), its magnitude is the column residual rc; its ROW index is
+  // round(rw / rc) where rw is the residual of the row-index-WEIGHTED
+  // column checksum.  This replaces the reference's row x column residual
+  // intersection (ft_sgemm_huge.cuh:422-485) whose row residuals needed a
+  // full cross-lane butterfly per accumulator register — the measured
+  // window cost dropped ~10x because no row reductions exist at all.
   auto locate_correct = [&]() __attribute__((always_inline)) {
-    // Column residuals: acc column == lane's r, so rc is lane-local.
-    float rc[FN];
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
-      float colp = 0.f;
+      float colp = 0.f, colw = 0.f;
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
-        for (int reg = 0; reg < NREG; ++reg) colp += acc[fm][fn][reg];
-      rc[fn] = slice_sum<MM>(colp) - slice_sum<MM>(cc[fn]);
-    }
-    // Row residuals and correction, one accumulator register at a time so
-    // the cold path holds only scalars (an array version here is pushed to
-    // scratch by the register allocator and its spill traffic poisons the
-    // hot loop — measured 2.5x).  Per reg: sum over the wave's columns
-    // (butterfly), fetch the row checksum from the lane that owns it
-    // (lane index == row index in sub-group 0), correct at row x column
-    // residual intersections (reference: ft_sgemm_huge.cuh:422-485; sign:
-    // residual = computed - checksum = +error, so subtract).
-#pragma unroll
-    for (int fm = 0; fm < FM; ++fm) {
-      const float crf = slice_sum<MM>(cr[fm]);  // full checksum for row r
-#pragma unroll
-      for (int reg = 0; reg < NREG; ++reg) {
-        float v = acc[fm][0][reg];
-#pragma unroll
-        for (int fn = 1; fn < FN; ++fn) v += acc[fm][fn][reg];
-        v = group_sum<MM>(v);
-        const float rr = v - __shfl(crf, acc_row(reg, sub), 64);
-        const bool rbad = fabsf(rr) > tau;
-#pragma unroll
-        for (int fn = 0; fn < FN; ++fn) {
-          const bool bad = rbad && (fabsf(rc[fn]) > tau);
-          acc[fm][fn][reg] -= bad ? rr : 0.f;
+        for (int reg = 0; reg < NREG; ++reg) {
+          const float v = acc[fm][fn][reg];
+          colp += v;
+          colw = fmaf((float)(fm * MM + acc_row(reg, sub)), v, colw);
         }
-      }
+      const float rc = slice_sum<MM>(colp) - slice_sum<MM>(cc[fn]);
+      const float rw = slice_sum<MM>(colw) - slice_sum<MM>(cw[fn]);
+      const bool cbad = fabsf(rc) > tau;
+      const int row = (int)rintf(rw / (cbad ? rc : 1.f));
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int reg = 0; reg < NREG; ++reg) {
+          const bool hit = cbad && (fm * MM + acc_row(reg, sub) == row);
+          acc[fm][fn][reg] -= hit ? rc : 0.f;
+        }
     }
   };
 
@@ -260,13 +272,13 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
     // Cheap detect: total tile sum vs total checksum (~70 VALU + 6
     // shuffles).  The residual is identical across the wave's lanes, so
     // the branch is uniform; only a wave that actually absorbed a fault
-    // enters the full locate/correct path above.
+    // enters the locate/correct path above.
     float tot = 0.f, chk = 0.f;
 #pragma unroll
-    for (int fm = 0; fm < FM; ++fm) {
-      chk += cr[fm];
+    for (int fn = 0; fn < FN; ++fn) {
+      chk += cc[fn];
 #pragma unroll
-      for (int fn = 0; fn < FN; ++fn)
+      for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
         for (int reg = 0; reg < NREG; ++reg) tot += acc[fm][fn][reg];
     }
@@ -328,15 +340,17 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
 
         if constexpr (ABFT) {
           // Encode: two broadcast LDS reads of the precomputed segment
-          // sums for this k-slice + one fma per fragment into the running
-          // checksums (reference encode: ft_sgemm_huge.cuh:150-213,
-          // redesigned around the offline segsum pass).
+          // sums (plain + row-weighted) for this k-slice + two fmas per
+          // B fragment into the running column checksums (reference
+          // encode: ft_sgemm_huge.cuh:150-213, redesigned around the
+          // offline segsum pass + ratio locate).
           const float sa = strip[kloc];
-          const float sb = strip[64 + kloc];
+          const float saw = strip[64 + kloc];
 #pragma unroll
-          for (int fm = 0; fm < FM; ++fm) cr[fm] = fmaf(a[fm], sb, cr[fm]);
-#pragma unroll
-          for (int fn = 0; fn < FN; ++fn) cc[fn] = fmaf(sa, b[fn], cc[fn]);
+          for (int fn = 0; fn < FN; ++fn) {
+            cc[fn] = fmaf(sa, b[fn], cc[fn]);
+            cw[fn] = fmaf(saw, b[fn], cw[fn]);
+          }
         }
 
 #pragma unroll

@@ -18,7 +18,9 @@ inline int abft_sstr(int K) { return (K + 63) & ~63; }
 
 template <int WM, int WN>
 size_t abft_workspace_floats_t(int M, int N, int K) {
-  return ((size_t)(M / WM) + (size_t)(N / WN)) * abft_sstr(K);
+  // plain + row-weighted A-segment sums (B-side sums are not needed by the
+  // ratio-locate scheme)
+  return 2 * (size_t)(M / WM) * abft_sstr(K);
 }
 
 template <int BM, int BN, int BK, int WM, int WN, int MM>
@@ -37,36 +39,30 @@ hipError_t launch_tier(bool abft, bool inject, int M, int N, int K,
                     : niter;
   if (stride < 1) stride = 1;
   const float* SA = nullptr;
-  const float* SB = nullptr;
   int sstr = 0;
   if (abft) {
-    // Precompute the segment checksums of A (per WM-row band) and B (per
-    // WN-col band): one coalesced pass over each operand, ~2% of the GEMM
+    // Precompute the plain + row-weighted segment checksums of A (per
+    // WM-row band): one coalesced pass over the operand, ~1-2% of the GEMM
     // at N=4096 (vs -11.6% for the in-kernel cooperative sums pass this
     // replaces — tools/probe_ablate.hip).
     if (!ws) return hipErrorInvalidValue;
     sstr = abft_sstr(K);
-    float* wsa = ws;
-    float* wsb = ws + (size_t)(M / WM) * sstr;
     hipLaunchKernelGGL((segsum_kernel<WM>), dim3(K), dim3(256), 0, stream, M,
-                       K, sstr, A, wsa);
-    hipLaunchKernelGGL((segsum_kernel<WN>), dim3(K), dim3(256), 0, stream, N,
-                       K, sstr, B, wsb);
-    SA = wsa;
-    SB = wsb;
+                       K, sstr, A, ws, ws + (size_t)(M / WM) * sstr);
+    SA = ws;
   }
   if (abft && inject) {
     hipLaunchKernelGGL((sgemm_mfma<BM, BN, BK, WM, WN, MM, true, true>), grid,
                        block, 0, stream, M, N, K, A, B, C, alpha, beta,
-                       stride, stride, tau, inj_mag, SA, SB, sstr);
+                       stride, stride, tau, inj_mag, SA, sstr);
   } else if (abft) {
     hipLaunchKernelGGL((sgemm_mfma<BM, BN, BK, WM, WN, MM, true, false>),
                        grid, block, 0, stream, M, N, K, A, B, C, alpha, beta,
-                       stride, stride, tau, inj_mag, SA, SB, sstr);
+                       stride, stride, tau, inj_mag, SA, sstr);
   } else {
     hipLaunchKernelGGL((sgemm_mfma<BM, BN, BK, WM, WN, MM, false, false>),
                        grid, block, 0, stream, M, N, K, A, B, C, alpha, beta,
-                       stride, stride, tau, inj_mag, SA, SB, sstr);
+                       stride, stride, tau, inj_mag, SA, sstr);
   }
   return hipGetLastError();
 }

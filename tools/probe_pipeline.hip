@@ -218,7 +218,7 @@ static void run_lib(const char* name, int n, const float* dA, const float* dB,
     hipLaunchKernelGGL(
         (ftsgemm::sgemm_mfma<BM, BN, 32, WM_, WN_, 32, false, false>), grid,
         block, 0, 0, n, n, n, dA, dB, dC, 1.f, 0.f, 128, 128, 1e30f, 0.f,
-        nullptr, nullptr, 0);
+        nullptr, 0);
   };
   hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
   launch();
