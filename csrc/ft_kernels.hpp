@@ -129,9 +129,12 @@ __global__ __launch_bounds__(256) void segsum_kernel(
   }
 }
 
+// NTC: non-temporal C epilogue loads/stores (beta != 0 streams 2x64 MB of C
+// through the caches per GEMM at N=4096 — nt keeps it from evicting the
+// L3-resident A/B panels).  SWIZ: bijective XCD-aware blockIdx remap.
 template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
-          bool INJECT>
-__global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
+          bool INJECT, bool NTC = false, bool SWIZ = false, int OCC = 2>
+__global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
     float beta, int verify_iters, int inject_stride, float tau, float inj_mag,
@@ -160,8 +163,22 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   const int wm_idx = wave / WAVES_N, wn_idx = wave % WAVES_N;
   const int wi0 = wm_idx * WM;
   const int wj0 = wn_idx * WN;
-  const int im0 = blockIdx.x * BM;
-  const int jn0 = blockIdx.y * BN;
+  int bx = blockIdx.x, by = blockIdx.y;
+  if constexpr (SWIZ) {
+    // bijective XCD remap: consecutive remapped ids land on one XCD so
+    // each XCD's L2 sees a contiguous band of output tiles.
+    const int nwg = gridDim.x * gridDim.y;
+    const int orig = by * gridDim.x + bx;
+    const int qq = nwg / 8, rr = nwg % 8;
+    const int xcd = orig % 8;
+    const int wgid =
+        (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) +
+        orig / 8;
+    bx = wgid % gridDim.x;
+    by = wgid / gridDim.x;
+  }
+  const int im0 = bx * BM;
+  const int jn0 = by * BN;
 
   typename T::acc_t acc[FM][FN] = {};
   float cc[FN] = {};  // running column checksum of this wave's tile (per
@@ -212,7 +229,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   // Workspace layout (set up by the launcher): SA = [plain A-segment sums
   // (M/WM rows) | row-weighted A-segment sums (M/WM rows)]; each row sstr
   // floats.  SB is unused by this scheme and passed null.
-  const int segA = blockIdx.x * WAVES_M + wm_idx;
+  const int segA = bx * WAVES_M + wm_idx;
   const int segsA = ABFT ? (M / WM) : 0;
   auto strip_stage = [&](int pb, int k0) __attribute__((always_inline)) {
     // glds source addresses are PER-LANE (the LDS side is uniform base +
@@ -243,6 +260,14 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   // full cross-lane butterfly per accumulator register — the measured
   // window cost dropped ~10x because no row reductions exist at all.
   auto locate_correct = [&]() __attribute__((always_inline)) {
+    // Opaque copy of `sub`: everything derived from it (the 2*FM*NREG
+    // per-lane row weights below) is materialised INSIDE this cold block.
+    // Without it the compiler hoists the loop-invariant weights into the
+    // kernel prologue and keeps them live through the whole hot loop
+    // (+68 VGPRs measured on the 64x64-wave tile -> spills on the bigger
+    // tiles).
+    int sub_o = sub;
+    asm volatile("" : "+v"(sub_o));
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
       float colp = 0.f, colw = 0.f;
@@ -252,7 +277,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
         for (int reg = 0; reg < NREG; ++reg) {
           const float v = acc[fm][fn][reg];
           colp += v;
-          colw = fmaf((float)(fm * MM + acc_row(reg, sub)), v, colw);
+          colw = fmaf((float)(fm * MM + acc_row(reg, sub_o)), v, colw);
         }
       const float rc = slice_sum<MM>(colp) - slice_sum<MM>(cc[fn]);
       const float rw = slice_sum<MM>(colw) - slice_sum<MM>(cw[fn]);
@@ -262,7 +287,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
         for (int reg = 0; reg < NREG; ++reg) {
-          const bool hit = cbad && (fm * MM + acc_row(reg, sub) == row);
+          const bool hit = cbad && (fm * MM + acc_row(reg, sub_o) == row);
           acc[fm][fn][reg] -= hit ? rc : 0.f;
         }
     }
@@ -295,8 +320,12 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
   // burst boundaries only (a conditional VALU read of acc inside the panel
   // loop makes hipcc shuttle all 64 accumulator registers AGPR<->VGPR every
   // panel — measured 2x wall time on the huge tier).
+  // A strip load covers 64 k values = PPS panels; strips are double
+  // buffered by strip-window parity.
+  constexpr int PPS = 64 / BK;  // panels per strip window (BK <= 64)
+  static_assert(64 % BK == 0 || !ABFT, "ABFT needs BK dividing 64");
   stage(0, 0);
-  if constexpr (ABFT) strip_stage(0, 0);  // pair (panels 0,1)
+  if constexpr (ABFT) strip_stage(0, 0);  // window (panels 0..PPS-1)
   __syncthreads();  // drains the in-flight glds (vmcnt(0) inside)
 
   const int niter = K / BK;
@@ -315,17 +344,17 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
       const int q = it & 1;
       if (it + 1 < niter) {
         stage(q ^ 1, (it + 1) * BK);
-        // a new strip pair rides along every other panel prefetch
+        // a new strip window rides along every PPS-th panel prefetch
         if constexpr (ABFT) {
-          if (((it + 1) & 1) == 0) strip_stage(((it + 1) >> 1) & 1,
+          if ((it + 1) % PPS == 0) strip_stage(((it + 1) / PPS) & 1,
                                                (it + 1) * BK);
         }
       }
       const float* As = &lds[q * BUF];
       const float* Bs = &lds[q * BUF + BM * BK];
       const float* strip =
-          ABFT ? &lds[STRIP_OFF + wave * 256 + ((it >> 1) & 1) * 128 +
-                      (it & 1) * 32]
+          ABFT ? &lds[STRIP_OFF + wave * 256 + ((it / PPS) & 1) * 128 +
+                      (it % PPS) * BK]
                : nullptr;
 #pragma unroll
       for (int kk = 0; kk < BK / KSTEP; ++kk) {
@@ -377,7 +406,8 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
         float* p = colbase + 4 * sub + 8 * g;
         f32x4 out;
         if (beta != 0.f) {
-          const f32x4 prev = *(const f32x4*)p;
+          const f32x4 prev = NTC ? __builtin_nontemporal_load((const f32x4*)p)
+                                 : *(const f32x4*)p;
 #pragma unroll
           for (int u = 0; u < 4; ++u)
             out[u] = alpha * acc[fm][fn][4 * g + u] + beta * prev[u];
@@ -385,7 +415,10 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), 2) void sgemm_mfma(
 #pragma unroll
           for (int u = 0; u < 4; ++u) out[u] = alpha * acc[fm][fn][4 * g + u];
         }
-        *(f32x4*)p = out;
+        if constexpr (NTC)
+          __builtin_nontemporal_store(out, (f32x4*)p);
+        else
+          *(f32x4*)p = out;
       }
     }
 }

@@ -30,7 +30,10 @@ TILING = OrderedDict(
     large=dict(bm=64, bn=64, bk=32, wm=64, wn=64, mfma="f32_32x32x2"),
     tall=dict(bm=128, bn=32, bk=32, wm=64, wn=32, mfma="f32_32x32x2"),
     wide=dict(bm=32, bn=128, bk=32, wm=32, wn=64, mfma="f32_32x32x2"),
-    huge=dict(bm=128, bn=128, bk=32, wm=64, wn=64, mfma="f32_32x32x2"),
+    # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32
+    # at N=4096 (tools/probe_pipeline.hip T3/T6 vs PA) — bigger M-tile cuts
+    # total A/B traffic 25% and the 48 KB LDS keeps 2 blocks/CU resident.
+    huge=dict(bm=256, bn=128, bk=16, wm=128, wn=64, mfma="f32_32x32x2"),
 )
 
 TIERS = list(TILING.keys())

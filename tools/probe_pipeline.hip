@@ -129,6 +129,51 @@ __global__ __launch_bounds__(256) void k_pipe(int M, int N, int K,
       kloop(q);
       __syncthreads();
     }
+  } else if constexpr (STRUCT == 3) {
+    // Register-pipelined: fragments for kk+1 load while kk computes, and the
+    // FIRST fragment of the next panel loads BEFORE the barrier (its buffer
+    // already landed — vmcnt-drained), so every wave has MFMA-ready
+    // operands the moment the barrier releases.  Raw s_barrier, no waits:
+    // all reads of the outgoing buffer were consumed by MFMAs (lgkm waits
+    // already emitted); the outstanding prefetch targets the incoming
+    // buffer, which nobody overwrites for another two panels.
+    auto frag_load = [&](const float* As, const float* Bs, int kk, float* af,
+                         float* bf) __attribute__((always_inline)) {
+      const int kloc = kk * 2 + sub;
+      af[0] = As[kloc * BM + wi0 + r];
+      af[1] = As[kloc * BM + wi0 + 32 + r];
+      bf[0] = Bs[kloc * BN + wj0 + r];
+      bf[1] = Bs[kloc * BN + wj0 + 32 + r];
+    };
+    stage(0, 0);
+    __syncthreads();
+    float af[2][2], bf[2][2];
+    frag_load(&lds[0], &lds[0] + BM * BK, 0, af[0], bf[0]);
+    for (int it = 0; it < niter; ++it) {
+      const int q = it & 1;
+      const float* As = &lds[q * BUF];
+      const float* Bs = As + BM * BK;
+      const float* An = &lds[(q ^ 1) * BUF];
+      const float* Bn = An + BM * BK;
+      if (it + 1 < niter) stage(q ^ 1, (it + 1) * BK);
+#pragma unroll
+      for (int kk = 0; kk < BK / 2; ++kk) {
+        const int cur = kk & 1, nxt = cur ^ 1;
+        if (kk + 1 < BK / 2) {
+          frag_load(As, Bs, kk + 1, af[nxt], bf[nxt]);
+        } else if (it + 1 < niter) {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+          frag_load(An, Bn, 0, af[nxt], bf[nxt]);
+        }
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+                af[cur][fm], bf[cur][fn], acc[fm][fn], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();
+    }
   } else if constexpr (STRUCT == 1) {
     // 2-buffer span: barrier first (frees q^1), issue next stage, THEN a
     // counted wait that drains only panel it — panel it+1 stays in flight
@@ -209,24 +254,45 @@ __global__ void max_diff(const float* x, const float* y, size_t n,
     }                                                               \
   } while (0)
 
-// Library-kernel wave-shape variants (WM x WN per wave; plain, no ABFT).
-template <int WM_, int WN_>
+// Library-kernel tile/wave-shape variants (plain or fused-ABFT+inject).
+template <int BM_, int BN_, int BK_, int WM_, int WN_, bool ABFT_ = false,
+          bool INJ_ = false, bool NTC_ = false, bool SWIZ_ = false,
+          int BETA10 = 0, int MM_ = 32, int OCC_ = 2>
 static void run_lib(const char* name, int n, const float* dA, const float* dB,
                     float* dC, const float* dRef, float* dMax, int reps) {
-  dim3 grid(n / BM, n / BN), block(64 * (BM / WM_) * (BN / WN_));
+  const float beta_ = BETA10 / 10.f;
+  dim3 grid(n / BM_, n / BN_), block(64 * (BM_ / WM_) * (BN_ / WN_));
+  static float* ws = nullptr;
+  const int sstr = (n + 63) & ~63;
+  if (ABFT_ && !ws) hipMalloc(&ws, 2 * (size_t)(8192 / 64) * 8192 * 4);
+  const int niter = n / BK_, stride = niter / 20 > 0 ? niter / 20 : 1;
   auto launch = [&]() {
-    hipLaunchKernelGGL(
-        (ftsgemm::sgemm_mfma<BM, BN, 32, WM_, WN_, 32, false, false>), grid,
-        block, 0, 0, n, n, n, dA, dB, dC, 1.f, 0.f, 128, 128, 1e30f, 0.f,
-        nullptr, 0);
+    if (ABFT_) {
+      hipLaunchKernelGGL((ftsgemm::segsum_kernel<WM_>), dim3(n), dim3(256),
+                         0, 0, n, n, sstr, dA, ws,
+                         ws + (size_t)(n / WM_) * sstr);
+      hipLaunchKernelGGL(
+          (ftsgemm::sgemm_mfma<BM_, BN_, BK_, WM_, WN_, MM_, ABFT_, INJ_,
+                               NTC_, SWIZ_, OCC_>),
+          grid, block, 0, 0, n, n, n, dA, dB, dC, 1.f, beta_, stride, stride,
+          9500.f, 10000.f, ws, sstr);
+    } else {
+      hipLaunchKernelGGL(
+          (ftsgemm::sgemm_mfma<BM_, BN_, BK_, WM_, WN_, MM_, false, false,
+                               NTC_, SWIZ_, OCC_>),
+          grid, block, 0, 0, n, n, n, dA, dB, dC, 1.f, beta_, stride, stride,
+          1e30f, 0.f, nullptr, 0);
+    }
   };
   hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
   launch();
   float md = 0.f;
-  hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
-  hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256), dim3(256),
-                     0, 0, dC, dRef, (size_t)n * n, dMax);
-  hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+  if (BETA10 == 0) {
+    hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+    hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256), dim3(256),
+                       0, 0, dC, dRef, (size_t)n * n, dMax);
+    hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+  }
   hipEvent_t b0, b1;
   hipEventCreate(&b0);
   hipEventCreate(&b1);
@@ -309,16 +375,22 @@ int main(int argc, char** argv) {
     for (int round = 0; round < 2; ++round) {
       printf("--- N=%d round %d\n", n, round);
       if (only && *only == 'L') {
-        run_lib<64, 128>("P8 lib 2-wave WM64xWN128", n, dA, dB, dC, dRef,
-                         dMax, reps);
-        run_lib<128, 64>("P9 lib 2-wave WM128xWN64", n, dA, dB, dC, dRef,
-                         dMax, reps);
-        run_lib<64, 64>("PA lib 4-wave (shipped)", n, dA, dB, dC, dRef, dMax,
-                        reps);
+        run_lib<128, 128, 32, 64, 64>("PA lib shipped", n, dA, dB, dC, dRef,
+                                      dMax, reps);
         continue;
       }
       run_variant<32, 0, false>("P0 bk32 2buf syncthreads", n, dA, dB, dC,
                                 dRef, dMax, reps);
+      run_variant<32, 3, false>("P10 bk32 regpipe x-barrier", n, dA, dB, dC,
+                                dRef, dMax, reps);
+      run_variant<16, 3, false>("P11 bk16 regpipe x-barrier", n, dA, dB, dC,
+                                dRef, dMax, reps);
+      run_variant<64, 3, false>("P12 bk64 regpipe x-barrier", n, dA, dB, dC,
+                                dRef, dMax, reps);
+      run_variant<16, 0, false>("PB bk16 2buf syncthreads", n, dA, dB, dC,
+                                dRef, dMax, reps);
+      run_variant<16, 1, false>("PC bk16 2buf rawbar/top-wait", n, dA, dB,
+                                dC, dRef, dMax, reps);
       run_variant<32, 1, false>("P1 bk32 2buf rawbar/top-wait", n, dA, dB, dC,
                                 dRef, dMax, reps);
       run_variant<64, 0, false>("P2 bk64 2buf syncthreads", n, dA, dB, dC,
@@ -333,12 +405,69 @@ int main(int argc, char** argv) {
                                dMax, reps);
       run_variant<32, 1, false, true>("P7 = P1 + setprio(1) on MFMA", n, dA,
                                       dB, dC, dRef, dMax, reps);
-      run_lib<64, 128>("P8 lib 2-wave WM64xWN128", n, dA, dB, dC, dRef, dMax,
-                       reps);
-      run_lib<128, 64>("P9 lib 2-wave WM128xWN64", n, dA, dB, dC, dRef, dMax,
-                       reps);
-      run_lib<64, 64>("PA lib 4-wave (shipped)", n, dA, dB, dC, dRef, dMax,
-                      reps);
+      run_lib<128, 128, 32, 64, 128>("P8 lib 128x128x32 w2 WM64xWN128", n,
+                                     dA, dB, dC, dRef, dMax, reps);
+      run_lib<128, 128, 32, 128, 64>("P9 lib 128x128x32 w2 WM128xWN64", n,
+                                     dA, dB, dC, dRef, dMax, reps);
+      run_lib<128, 128, 32, 64, 64>("PA lib 128x128x32 w4 (shipped)", n, dA,
+                                    dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 32, 64, 128>("T1 lib 256x128x32 w4 WM64xWN128", n,
+                                     dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 64, 128>("T2 lib 256x128x16 w4 WM64xWN128", n,
+                                     dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64>("T3 lib 256x128x16 w4 WM128xWN64", n,
+                                     dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 8, 128, 64>("T6 lib 256x128x8 w4 WM128xWN64", n, dA,
+                                    dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 32, 64, 64>("T7 lib 256x128x32 w8 WM64xWN64", n, dA,
+                                    dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 64, 64>("T8 lib 256x128x16 w8 WM64xWN64", n, dA,
+                                    dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, true, true>(
+          "A1 abft+inj 256x128x16 w4 WM128xWN64", n, dA, dB, dC, dRef, dMax,
+          reps);
+      run_lib<256, 128, 16, 64, 128, true, true>(
+          "A2 abft+inj 256x128x16 w4 WM64xWN128", n, dA, dB, dC, dRef, dMax,
+          reps);
+      run_lib<256, 128, 16, 64, 64, true, true>(
+          "A3 abft+inj 256x128x16 w8 WM64xWN64", n, dA, dB, dC, dRef, dMax,
+          reps);
+      run_lib<128, 128, 16, 64, 64, true, true>(
+          "A4 abft+inj 128x128x16 w4 WM64xWN64", n, dA, dB, dC, dRef, dMax,
+          reps);
+      run_lib<128, 128, 32, 64, 64, true, true>(
+          "A5 abft+inj 128x128x32 w4 WM64xWN64", n, dA, dB, dC, dRef, dMax,
+          reps);
+      run_lib<256, 128, 16, 64, 64>("A3p plain  256x128x16 w8 WM64xWN64", n,
+                                    dA, dB, dC, dRef, dMax, reps);
+      // beta = -1.5 family (C read-modify-write traffic): NTC / SWIZ
+      run_lib<256, 128, 16, 128, 64, false, false, false, false, -15>(
+          "B0 plain b-1.5 base", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, true, false, -15>(
+          "B1 plain b-1.5 +NTC", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, false, true, -15>(
+          "B2 plain b-1.5 +SWIZ", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, true, true, -15>(
+          "B3 plain b-1.5 +NTC+SWIZ", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, true, true, true, true, -15>(
+          "B4 abft+inj b-1.5 +NTC+SWIZ", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, true, true, false, false, -15>(
+          "B5 abft+inj b-1.5 base", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, true, true, 0>(
+          "B6 plain b0 +NTC+SWIZ", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, false, false, -15, 16>(
+          "C1 plain b-1.5 MM16 (16x16x4)", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, false, false, -15, 32, 3>(
+          "C2 plain b-1.5 bk16 OCC3", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 8, 128, 64, false, false, false, false, -15, 32, 3>(
+          "C3 plain b-1.5 bk8 OCC3", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 8, 128, 64, false, false, false, false, -15, 32, 2>(
+          "C4 plain b-1.5 bk8 OCC2", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<128, 128, 16, 128, 64, false, false, false, false, -15, 32, 4>(
+          "C5 plain b-1.5 128x128x16 w2 OCC4", n, dA, dB, dC, dRef, dMax,
+          reps);
+      run_lib<256, 128, 16, 128, 64, true, true, false, false, -15, 32, 3>(
+          "C6 abft+inj b-1.5 bk16 OCC3", n, dA, dB, dC, dRef, dMax, reps);
     }
     hipFree(dA); hipFree(dB); hipFree(dC); hipFree(dRef); hipFree(dMax);
   }
