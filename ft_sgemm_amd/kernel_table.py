@@ -24,12 +24,16 @@ Fields per tier:
 
 from collections import OrderedDict
 
+# BK=16 on the mid tiers: measured +8..+19% plain and up to +23% fused at
+# N=2048..4096 (tools/probe_pipeline.hip PROBE_ONLY=T): halving the LDS
+# panel keeps one more block resident per CU, which covers the per-panel
+# barrier park.
 TILING = OrderedDict(
     small=dict(bm=16, bn=16, bk=32, wm=16, wn=16, mfma="f32_16x16x4"),
-    medium=dict(bm=32, bn=32, bk=32, wm=32, wn=32, mfma="f32_32x32x2"),
-    large=dict(bm=64, bn=64, bk=32, wm=64, wn=64, mfma="f32_32x32x2"),
-    tall=dict(bm=128, bn=32, bk=32, wm=64, wn=32, mfma="f32_32x32x2"),
-    wide=dict(bm=32, bn=128, bk=32, wm=32, wn=64, mfma="f32_32x32x2"),
+    medium=dict(bm=32, bn=32, bk=16, wm=32, wn=32, mfma="f32_32x32x2"),
+    large=dict(bm=64, bn=64, bk=16, wm=64, wn=64, mfma="f32_32x32x2"),
+    tall=dict(bm=128, bn=32, bk=16, wm=64, wn=32, mfma="f32_32x32x2"),
+    wide=dict(bm=32, bn=128, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
     # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32
     # at N=4096 (tools/probe_pipeline.hip T3/T6 vs PA) — bigger M-tile cuts
     # total A/B traffic 25% and the 48 KB LDS keeps 2 blocks/CU resident.

@@ -379,6 +379,46 @@ int main(int argc, char** argv) {
                                       dMax, reps);
         continue;
       }
+      if (only && *only == 'T') {
+        // tier tuning: tall / wide / large variants (beta=-1.5)
+        run_lib<128, 32, 32, 64, 32, false, false, false, false, -15>(
+            "tall    bk32 plain (shipped)", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 32, 16, 64, 32, false, false, false, false, -15>(
+            "tall    bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 32, 32, 64, 32, true, true, false, false, -15>(
+            "tall    bk32 abft+inj (shipped)", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<128, 32, 16, 64, 32, true, true, false, false, -15>(
+            "tall    bk16 abft+inj", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 32, 16, 128, 32, true, true, false, false, -15>(
+            "tall    bk16 w1 WM128 abft+inj", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<32, 128, 32, 32, 64, false, false, false, false, -15>(
+            "wide    bk32 plain (shipped)", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 128, 16, 32, 64, false, false, false, false, -15>(
+            "wide    bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 128, 16, 32, 64, true, true, false, false, -15>(
+            "wide    bk16 abft+inj", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 128, 16, 32, 128, true, true, false, false, -15>(
+            "wide    bk16 w1 WN128 abft+inj", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<64, 64, 32, 64, 64, false, false, false, false, -15>(
+            "large   bk32 plain (shipped)", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 64, 16, 64, 64, false, false, false, false, -15>(
+            "large   bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 64, 16, 64, 64, false, false, false, false, -15, 32, 4>(
+            "large   bk16 occ4 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 64, 16, 64, 64, true, true, false, false, -15>(
+            "large   bk16 abft+inj", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 64, 32, 64, 64, true, true, false, false, -15>(
+            "large   bk32 abft+inj (shipped)", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<32, 32, 16, 32, 32, false, false, false, false, -15>(
+            "medium  bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 32, 16, 32, 32, true, true, false, false, -15>(
+            "medium  bk16 abft+inj", n, dA, dB, dC, dRef, dMax, reps);
+        continue;
+      }
       run_variant<32, 0, false>("P0 bk32 2buf syncthreads", n, dA, dB, dC,
                                 dRef, dMax, reps);
       run_variant<32, 3, false>("P10 bk32 regpipe x-barrier", n, dA, dB, dC,
