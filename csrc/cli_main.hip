@@ -75,9 +75,16 @@ bool run_kernel(int kid, int M, int N, int K, const float* dA,
                                       w, 0) == hipSuccess;
   }
   if (kid == 10) {
+    // panel width sized for MI355X (the reference's 256 is T4-sized: at
+    // these GFLOPS the per-panel full-C gemv sweeps and launch overhead
+    // dominate; FT_SGEMM_PANEL_K overrides)
+    static const int panel_k = [] {
+      const char* e = getenv("FT_SGEMM_PANEL_K");
+      return e ? atoi(e) : 1024;
+    }();
     float r0 = 0, r1 = 0;
     return ftsgemm::baseline_ft_sgemm(M, N, K, dA, dB, dC, alpha, beta, ws,
-                                      256, &r0, &r1, 0) == 0;
+                                      panel_k, &r0, &r1, 0) == 0;
   }
   // ids 0, 7, 8, 9: rocBLAS (reference falls back to cuBLAS, sgemm.cu:197)
   return ftsgemm::rocblas_sgemm_nt(M, N, K, dA, dB, dC, alpha, beta, 0) == 0;
@@ -120,21 +127,23 @@ int main(int argc, char** argv) {
   HIP_CALL(hipMemcpy(dB, hB.data(), bytes, hipMemcpyHostToDevice));
 
   // Baseline (id 10) checksum workspace
-  float *ws_ones, *ws_rowc, *ws_colc, *ws_sa, *ws_sb, *ws_rrow, *ws_rcol;
+  float *ws_ones, *ws_rowc, *ws_colc, *ws_sa, *ws_sb, *ws_rrow, *ws_rcol,
+      *ws_dres;
   HIP_CALL(hipMalloc(&ws_ones, maxn * sizeof(float)));
   HIP_CALL(hipMalloc(&ws_rowc, maxn * sizeof(float)));
   HIP_CALL(hipMalloc(&ws_colc, maxn * sizeof(float)));
-  HIP_CALL(hipMalloc(&ws_sa, 256 * sizeof(float)));
-  HIP_CALL(hipMalloc(&ws_sb, 256 * sizeof(float)));
+  HIP_CALL(hipMalloc(&ws_sa, maxn * sizeof(float)));  // >= panel_k
+  HIP_CALL(hipMalloc(&ws_sb, maxn * sizeof(float)));
   HIP_CALL(hipMalloc(&ws_rrow, maxn * sizeof(float)));
   HIP_CALL(hipMalloc(&ws_rcol, maxn * sizeof(float)));
+  HIP_CALL(hipMalloc(&ws_dres, 2 * sizeof(float)));
   {
     std::vector<float> ones(maxn, 1.f);
     HIP_CALL(hipMemcpy(ws_ones, ones.data(), maxn * sizeof(float),
                        hipMemcpyHostToDevice));
   }
-  ftsgemm::BaselineWorkspace ws{ws_ones, ws_rowc, ws_colc,
-                                ws_sa,   ws_sb,   ws_rrow, ws_rcol};
+  ftsgemm::BaselineWorkspace ws{ws_ones, ws_rowc, ws_colc, ws_sa,
+                                ws_sb,   ws_rrow, ws_rcol, ws_dres};
 
   // ---------------- verification pass ----------------
   const int vM = start, vN = start, vK = start;

@@ -46,6 +46,11 @@ int baseline_ft_sgemm(int M, int N, int K, const float* A, const float* B,
                       float* res_row, float* res_col, hipStream_t stream) {
   rocblas_handle h = get_handle();
   RB_CHECK(rocblas_set_stream(h, stream));
+  // Host pointer mode for the alpha/beta scalars; the per-panel dot
+  // verdicts go to DEVICE memory (ws.d_res) so the chain stays fully
+  // stream-ordered — a host-pointer sdot would synchronize every panel
+  // (the reference pays that sync at baseline_ft_sgemm.cuh:28,31; on
+  // MI355X it costs more than the dot itself).
   RB_CHECK(rocblas_set_pointer_mode(h, rocblas_pointer_mode_host));
   const float one = 1.f, zero = 0.f, neg1 = -1.f;
 
@@ -82,12 +87,22 @@ int baseline_ft_sgemm(int M, int N, int K, const float* A, const float* B,
     RB_CHECK(rocblas_sgemv(h, rocblas_operation_transpose, M, N, &one, C, M,
                            ws.ones, 1, &zero, ws.col_c, 1));
     // Residual + scalar verdict (axpy + dot, as the reference does at
-    // baseline_ft_sgemm.cuh:25-31)
+    // baseline_ft_sgemm.cuh:25-31), verdicts stream-ordered into device
+    // memory
     RB_CHECK(rocblas_saxpy(h, M, &neg1, ws.ref_row, 1, ws.row_c, 1));
-    RB_CHECK(rocblas_sdot(h, M, ws.row_c, 1, ws.row_c, 1, res_row));
     RB_CHECK(rocblas_saxpy(h, N, &neg1, ws.ref_col, 1, ws.col_c, 1));
-    RB_CHECK(rocblas_sdot(h, N, ws.col_c, 1, ws.col_c, 1, res_col));
+    RB_CHECK(rocblas_set_pointer_mode(h, rocblas_pointer_mode_device));
+    RB_CHECK(rocblas_sdot(h, M, ws.row_c, 1, ws.row_c, 1, ws.d_res));
+    RB_CHECK(rocblas_sdot(h, N, ws.col_c, 1, ws.col_c, 1, ws.d_res + 1));
+    RB_CHECK(rocblas_set_pointer_mode(h, rocblas_pointer_mode_host));
   }
+  float res[2];
+  if (hipMemcpyAsync(res, ws.d_res, 2 * sizeof(float),
+                     hipMemcpyDeviceToHost, stream) != hipSuccess ||
+      hipStreamSynchronize(stream) != hipSuccess)
+    return -1;
+  *res_row = res[0];
+  *res_col = res[1];
   return 0;
 }
 

@@ -80,11 +80,12 @@ std::tuple<double, double> baseline_ft(at::Tensor a, at::Tensor b,
   at::Tensor row_c = at::empty({M}, opts), col_c = at::empty({N}, opts);
   at::Tensor s_a = at::empty({panel_k}, opts), s_b = at::empty({panel_k}, opts);
   at::Tensor ref_row = at::empty({M}, opts), ref_col = at::empty({N}, opts);
+  at::Tensor d_res = at::empty({2}, opts);
   ftsgemm::BaselineWorkspace ws{
       ones.mutable_data_ptr<float>(),    row_c.mutable_data_ptr<float>(),
       col_c.mutable_data_ptr<float>(),   s_a.mutable_data_ptr<float>(),
       s_b.mutable_data_ptr<float>(),     ref_row.mutable_data_ptr<float>(),
-      ref_col.mutable_data_ptr<float>()};
+      ref_col.mutable_data_ptr<float>(), d_res.mutable_data_ptr<float>()};
   float res_row = 0.f, res_col = 0.f;
   auto stream = at::cuda::getCurrentCUDAStream();
   int st = ftsgemm::baseline_ft_sgemm(

@@ -66,7 +66,7 @@ def rocblas_sgemm(a, b, c, alpha: float = 1.0, beta: float = 0.0):
 
 
 def baseline_ft(a, b, c, alpha: float = 1.0, beta: float = 0.0,
-                panel_k: int = 256):
+                panel_k: int = 1024):
     """Kernel id 10: non-fused rocBLAS ABFT chain.  Returns (c, verdicts)."""
     res = _require_ext().baseline_ft(a, b, c, alpha, beta, panel_k)
     return c, res
