@@ -117,3 +117,21 @@ def test_native_extension_is_loaded():
     """Guards against silent eager fallback: the .so must be in-tree."""
     import ft_sgemm_amd._C as ext
     assert "ft_sgemm_amd" in ext.__file__
+
+
+def test_block_row_fused_abft_world1():
+    """The distributed block-row path (BASELINE configs[4b]) on one GPU:
+    K-panel loop accumulating with the fused-ABFT huge kernel + injection."""
+    _require_native()
+    from ft_sgemm_amd.parallel import block_row_sgemm
+    m = n = k = 512
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+
+    def gemm_fn(ap, bp, cl, al, be):
+        ops.ft_sgemm("huge", ap, bp, cl, al, be, inject=True)
+
+    block_row_sgemm(a, b, c, panel_k=256, gemm_fn=gemm_fn, alpha=1.0,
+                    beta=0.0)
+    torch.cuda.synchronize()
+    check(ref, c)
