@@ -103,6 +103,9 @@ int main(int argc, char** argv) {
   }
   const char* no_inj = getenv("FT_SGEMM_NO_INJECT");
   const bool inject = !(no_inj && strcmp(no_inj, "0") != 0);
+  const char* json_env = getenv("FT_SGEMM_JSON");  // machine-readable lines
+  FILE* jf = nullptr;
+  if (json_env && *json_env) jf = fopen(json_env, "w");
   const int reps = 5;
   const float alpha = 1.f;
   const size_t maxn = (size_t)end;
@@ -203,11 +206,18 @@ int main(int argc, char** argv) {
       HIP_CALL(hipEventElapsedTime(&ms, beg, fin));
       double gflops = 2.0 * M * N * K * reps / (ms * 1e-3) / 1e9;
       printf("|%8.0f", gflops);
+      if (jf)
+        fprintf(jf,
+                "{\"kernel_id\": %d, \"name\": \"%s\", \"n\": %d, "
+                "\"gflops\": %.1f, \"reps\": %d, \"inject\": %s}\n",
+                kid, kNames[kid], n, gflops, reps,
+                (kid >= 11 && inject) ? "true" : "false");
       HIP_CALL(hipEventDestroy(beg));
       HIP_CALL(hipEventDestroy(fin));
       fflush(stdout);
     }
     printf("|\n");
   }
+  if (jf) fclose(jf);
   return 0;
 }

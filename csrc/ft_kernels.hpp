@@ -132,8 +132,12 @@ __global__ __launch_bounds__(256) void segsum_kernel(
 // NTC: non-temporal C epilogue loads/stores (beta != 0 streams 2x64 MB of C
 // through the caches per GEMM at N=4096 — nt keeps it from evicting the
 // L3-resident A/B panels).  SWIZ: bijective XCD-aware blockIdx remap.
+// PIPE=1 (plain kernels only): 3-LDS-buffer glds ring with raw s_barrier +
+// counted s_waitcnt — panels it+1 and it+2 stay in flight across the
+// barrier instead of draining at every __syncthreads.
 template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
-          bool INJECT, bool NTC = false, bool SWIZ = false, int OCC = 2>
+          bool INJECT, bool NTC = false, bool SWIZ = false, int OCC = 2,
+          int PIPE = 0>
 __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
@@ -151,8 +155,11 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
   // saw(64)] window of the precomputed segment sums, streamed by
   // global_load_lds.  Only A-side sums are needed: the column checksums
   // detect, the weighted ones locate the row, and the column is lane-local.
-  constexpr int STRIP_OFF = 2 * BUF;
-  constexpr int LDS_FLOATS = ABFT ? (STRIP_OFF + NWAVES * 256) : (2 * BUF);
+  static_assert(!(ABFT && PIPE), "3-buffer ring is plain-only");
+  constexpr int NBUF = PIPE ? 3 : 2;
+  constexpr int STRIP_OFF = NBUF * BUF;
+  constexpr int LDS_FLOATS = ABFT ? (STRIP_OFF + NWAVES * 256)
+                                  : (NBUF * BUF);
 
   __shared__ __attribute__((aligned(16))) float lds[LDS_FLOATS];
 
@@ -324,12 +331,51 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
   // buffered by strip-window parity.
   constexpr int PPS = 64 / BK;  // panels per strip window (BK <= 64)
   static_assert(64 % BK == 0 || !ABFT, "ABFT needs BK dividing 64");
+  constexpr int GLD = GA + GB;  // glds instructions per panel
   stage(0, 0);
   if constexpr (ABFT) strip_stage(0, 0);  // window (panels 0..PPS-1)
-  __syncthreads();  // drains the in-flight glds (vmcnt(0) inside)
+  if constexpr (PIPE) {
+    // 3-buffer ring: prologue stages two panels; the main loop keeps the
+    // newest one in flight across each barrier.
+    if (BK < K) stage(1, BK);
+  }
+  if constexpr (!PIPE) __syncthreads();  // drains in-flight glds
 
   const int niter = K / BK;
-  int it = 0;
+  if constexpr (PIPE) {
+    for (int it = 0; it < niter; ++it) {
+      const int q = it % 3;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();  // readers of buf (it+2)%3 are done
+      if (it + 2 < niter) {
+        stage((it + 2) % 3, (it + 2) * BK);
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(2 * GLD) : "memory");
+      } else if (it + 1 < niter) {
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(GLD) : "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      const float* As = &lds[q * BUF];
+      const float* Bs = &lds[q * BUF + BM * BK];
+#pragma unroll
+      for (int kk = 0; kk < BK / KSTEP; ++kk) {
+        const int kloc = kk * KSTEP + sub;
+        float a[FM], b[FN];
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm)
+          a[fm] = As[kloc * BM + wi0 + fm * MM + r];
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn)
+          b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < FN; ++fn)
+            acc[fm][fn] = T::mma(a[fm], b[fn], acc[fm][fn]);
+      }
+    }
+  }
+  int it = PIPE ? niter : 0;  // PIPE path already done; skip burst loop
   while (it < niter) {
     if constexpr (INJECT) {
       // Deterministic rotating injector, once per verify window (reference:

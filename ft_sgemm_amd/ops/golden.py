@@ -126,6 +126,34 @@ def ft_sgemm_golden(a, b, c, alpha=1.0, beta=0.0, tau=ERR_BOUND,
     return out, injected_sites, located
 
 
+def abft_ratio_locate(prod: np.ndarray, a: np.ndarray, b: np.ndarray,
+                      seg: int, tau: float = ERR_BOUND):
+    """Golden model of the DEVICE locate scheme (csrc/ft_kernels.hpp):
+    per seg-row band of C, maintain per-column plain (cc) and
+    row-index-weighted (cw) checksums from the precomputed segment sums of
+    A; a fault's column has |rc| > tau, its magnitude is rc, and its row
+    within the band is round(rw / rc).  Returns (corrected, locations) with
+    locations in global (i, j) coordinates."""
+    m, n = prod.shape
+    out = prod.copy()
+    locs = []
+    w = np.arange(seg, dtype=np.float32)
+    for s0 in range(0, m, seg):
+        band = out[s0:s0 + seg]
+        ab = a[s0:s0 + seg].astype(np.float32)
+        sa = ab.sum(axis=0, dtype=np.float32)           # plain segment sums
+        saw = w @ ab                                    # weighted
+        cc = (b.astype(np.float32) @ sa).astype(np.float32)     # N
+        cw = (b.astype(np.float32) @ saw).astype(np.float32)    # N
+        rc = band.sum(axis=0, dtype=np.float32) - cc
+        rw = (w @ band).astype(np.float32) - cw
+        for j in np.nonzero(np.abs(rc) > tau)[0]:
+            i = int(np.rint(rw[j] / rc[j]))
+            band[i, j] -= rc[j]
+            locs.append((s0 + i, int(j)))
+    return out.astype(np.float32), locs
+
+
 def baseline_ft_check(a, b, panel_k: int = 256, tau: float = ERR_BOUND):
     """Non-fused baseline semantics (include/baseline_ft_sgemm.cuh:1-33):
     per 256-wide K panel, compare e^T(A@B^T) against (e^T A)@B^T and

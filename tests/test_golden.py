@@ -91,3 +91,40 @@ def test_threshold_no_false_positives_large_k():
     prod = (a @ b.T).astype(np.float32)
     _, locs = golden.abft_detect_correct(prod, cks)
     assert locs == []
+
+
+def test_ratio_locate_single_fault_per_band():
+    """Golden model of the device's ratio locate (plain + row-weighted
+    column checksums per 64-row band): exact location and correction."""
+    a, b, _ = make(m=256, n=192, k=320)
+    prod = (a @ b.T).astype(np.float32)
+    sites = [(17, 93), (64 + 3, 5), (128 + 63, 191), (192 + 31, 0)]
+    for i, j in sites:
+        prod[i, j] += np.float32(1e4)
+    corrected, locs = golden.abft_ratio_locate(prod, a, b, seg=64)
+    assert sorted(locs) == sorted(sites)
+    ref = a.astype(np.float64) @ b.astype(np.float64).T
+    ok, idx, _ = verify_matrix(ref, corrected)
+    assert ok, f"ratio correction left a mismatch at {idx}"
+
+
+def test_ratio_locate_matches_intersection_scheme():
+    """Both locate schemes agree on a clean single fault."""
+    a, b, _ = make()
+    prod = (a @ b.T).astype(np.float32)
+    prod[100, 20] += np.float32(1e4)
+    cks = golden.abft_encode(a, b)
+    c1, l1 = golden.abft_detect_correct(prod, cks)
+    c2, l2 = golden.abft_ratio_locate(prod, a, b, seg=64)
+    assert l1 == [(100, 20)]
+    assert l2 == [(100, 20)]
+    # corrections differ only by checksum roundoff paths
+    assert np.abs(c1 - c2).max() < 1e-2
+
+
+def test_ratio_locate_clean_no_false_positive():
+    a, b, _ = make(m=256, n=256, k=1024)
+    prod = (a @ b.T).astype(np.float32)
+    corrected, locs = golden.abft_ratio_locate(prod, a, b, seg=64)
+    assert locs == []
+    assert np.array_equal(corrected, prod)

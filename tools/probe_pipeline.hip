@@ -257,7 +257,7 @@ __global__ void max_diff(const float* x, const float* y, size_t n,
 // Library-kernel tile/wave-shape variants (plain or fused-ABFT+inject).
 template <int BM_, int BN_, int BK_, int WM_, int WN_, bool ABFT_ = false,
           bool INJ_ = false, bool NTC_ = false, bool SWIZ_ = false,
-          int BETA10 = 0, int MM_ = 32, int OCC_ = 2>
+          int BETA10 = 0, int MM_ = 32, int OCC_ = 2, int PIPE_ = 0>
 static void run_lib(const char* name, int n, const float* dA, const float* dB,
                     float* dC, const float* dRef, float* dMax, int reps) {
   const float beta_ = BETA10 / 10.f;
@@ -279,7 +279,7 @@ static void run_lib(const char* name, int n, const float* dA, const float* dB,
     } else {
       hipLaunchKernelGGL(
           (ftsgemm::sgemm_mfma<BM_, BN_, BK_, WM_, WN_, MM_, false, false,
-                               NTC_, SWIZ_, OCC_>),
+                               NTC_, SWIZ_, OCC_, PIPE_>),
           grid, block, 0, 0, n, n, n, dA, dB, dC, 1.f, beta_, stride, stride,
           1e30f, 0.f, nullptr, 0);
     }
@@ -417,6 +417,18 @@ int main(int argc, char** argv) {
             "medium  bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
         run_lib<32, 32, 16, 32, 32, true, true, false, false, -15>(
             "medium  bk16 abft+inj", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 32, 16, 64, 32, true, true, false, false, -15, 32, 3>(
+            "tall    bk16 abft+inj OCC3", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 32, 16, 64, 32, true, true, false, false, -15, 32, 4>(
+            "tall    bk16 abft+inj OCC4", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 128, 16, 32, 64, true, true, false, false, -15, 32, 3>(
+            "wide    bk16 abft+inj OCC3", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 128, 16, 32, 64, true, true, false, false, -15, 32, 4>(
+            "wide    bk16 abft+inj OCC4", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 64, 16, 64, 64, true, true, false, false, -15, 32, 3>(
+            "large   bk16 abft+inj OCC3", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<32, 32, 16, 32, 32, true, true, false, false, -15, 32, 4>(
+            "medium  bk16 abft+inj OCC4", n, dA, dB, dC, dRef, dMax, reps);
         continue;
       }
       run_variant<32, 0, false>("P0 bk32 2buf syncthreads", n, dA, dB, dC,
@@ -508,6 +520,15 @@ int main(int argc, char** argv) {
           reps);
       run_lib<256, 128, 16, 128, 64, true, true, false, false, -15, 32, 3>(
           "C6 abft+inj b-1.5 bk16 OCC3", n, dA, dB, dC, dRef, dMax, reps);
+      run_lib<256, 128, 16, 128, 64, false, false, false, false, -15, 32, 2,
+              1>("E1 plain b-1.5 bk16 3buf-ring", n, dA, dB, dC, dRef, dMax,
+                 reps);
+      run_lib<256, 128, 8, 128, 64, false, false, false, false, -15, 32, 2,
+              1>("E2 plain b-1.5 bk8 3buf-ring", n, dA, dB, dC, dRef, dMax,
+                 reps);
+      run_lib<256, 128, 32, 64, 128, false, false, false, false, -15, 32, 2,
+              1>("E3 plain b-1.5 bk32 3buf-ring w4 WM64WN128", n, dA, dB,
+                 dC, dRef, dMax, reps);
     }
     hipFree(dA); hipFree(dB); hipFree(dC); hipFree(dRef); hipFree(dMax);
   }
