@@ -36,17 +36,22 @@ def main():
                     choices=["abft_huge", "huge", "rocblas"])
     ap.add_argument("--mode", default="replicated",
                     choices=["replicated", "blockrow"])
+    ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
+                    help="cpu: CI smoke of the full multi-rank bench "
+                         "contract over gloo (tiny size, torch matmul)")
     args = ap.parse_args()
 
-    if not torch.cuda.is_available():
+    cpu_ci = args.device == "cpu"
+    if not cpu_ci and not torch.cuda.is_available():
         raise SystemExit("bench.py needs a ROCm GPU (run under gpurun)")
 
     from ft_sgemm_amd import ops
     from ft_sgemm_amd.parallel import init_from_env
 
-    rank, world = init_from_env()
+    rank, world = init_from_env(backend="gloo" if cpu_ci else None)
     import torch.distributed as dist
-    dev = torch.device("cuda", torch.cuda.current_device())
+    dev = (torch.device("cpu") if cpu_ci
+           else torch.device("cuda", torch.cuda.current_device()))
     n = args.size
     torch.manual_seed(10 + rank)
     a = (torch.rand((n, n), device=dev) * 1.8 - 0.9).contiguous()
@@ -64,9 +69,14 @@ def main():
         b_loc = b[:, : nhi - nlo].contiguous()   # (K, N_loc)
         c_loc = torch.zeros((n, mhi - mlo), device=dev)
         panel_k = max(1024, n // 16)
+        if n % panel_k or panel_k > n:
+            panel_k = n  # tiny / odd sizes: single panel
 
         def gemm_fn(ap, bp, cl, al, be):
-            if args.kernel == "abft_huge":
+            if cpu_ci:
+                from ft_sgemm_amd.parallel.distributed import torch_gemm_fn
+                torch_gemm_fn(ap, bp, cl, al, be)
+            elif args.kernel == "abft_huge":
                 ops.ft_sgemm("huge", ap, bp, cl, al, be, inject=True)
             elif args.kernel == "huge":
                 ops.sgemm("huge", ap, bp, cl, al, be)
@@ -75,6 +85,8 @@ def main():
 
         step = lambda: block_row_sgemm(a_loc, b_loc, c_loc, panel_k=panel_k,
                                        gemm_fn=gemm_fn, alpha=1.0, beta=-1.5)
+    elif cpu_ci:
+        step = lambda: torch.matmul(b.transpose(0, 1), a)
     elif args.kernel == "abft_huge":
         step = lambda: ops.ft_sgemm("huge", a, b, c, 1.0, -1.5, inject=True)
     elif args.kernel == "huge":
@@ -84,21 +96,25 @@ def main():
 
     for _ in range(args.warmup):
         step()
-    torch.cuda.synchronize()
+    if not cpu_ci:
+        torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
-        torch.cuda.synchronize()
+        if not cpu_ci:
+            torch.cuda.synchronize()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
-    torch.cuda.synchronize()
+    if not cpu_ci:
+        torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
     if world > 1:
         dist.barrier()
-        t = torch.tensor([elapsed], device=dev)
+        t = torch.tensor([elapsed], device="cpu" if cpu_ci else dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        torch.cuda.synchronize()
+        if not cpu_ci:
+            torch.cuda.synchronize()
         elapsed = float(t.item())
 
     # replicated: every rank computes an independent n^3 GEMM (weak);

@@ -82,3 +82,48 @@ def test_local_shard():
     assert local_shard(32768, 3, 8) == (3 * 4096, 4 * 4096)
     with pytest.raises(AssertionError):
         local_shard(100, 0, 3)
+
+
+def _bench_worker(rank, world, port, mode, q):
+    import io
+    import subprocess
+    import sys
+    env = dict(os.environ, RANK=str(rank), WORLD_SIZE=str(world),
+               LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+               MASTER_PORT=str(port))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--device", "cpu", "--size", "128",
+         "--steps", "3", "--warmup", "1", "--mode", mode],
+        capture_output=True, text=True, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    q.put((rank, r.returncode, r.stdout, r.stderr))
+
+
+@pytest.mark.parametrize("mode", ["replicated", "blockrow"])
+def test_bench_contract_world2_cpu(mode):
+    """End-to-end smoke of the driver's bench contract at world_size 2 over
+    gloo: rank 0 must print exactly one valid JSON line with the whole-job
+    aggregate, rank 1 nothing."""
+    import json
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29981 if mode == "replicated" else 29982
+    procs = [ctx.Process(target=_bench_worker, args=(r, 2, port, mode, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, rc, out, err = q.get()
+        results[rank] = (rc, out, err)
+    for p in procs:
+        p.join(timeout=180)
+    for rank, (rc, out, err) in results.items():
+        assert rc == 0, f"rank {rank} failed: {err[-800:]}"
+    payload = [l for l in results[0][1].splitlines() if l.startswith("{")]
+    assert len(payload) == 1, results[0][1]
+    j = json.loads(payload[0])
+    assert j["n_gpus"] == 2
+    assert j["metric"] == "fused_abft_sgemm_gflops"
+    assert j["scaling"] == ("weak" if mode == "replicated" else "strong")
+    assert not [l for l in results[1][1].splitlines() if l.startswith("{")]
