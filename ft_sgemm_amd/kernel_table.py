@@ -31,7 +31,9 @@ from collections import OrderedDict
 TILING = OrderedDict(
     small=dict(bm=16, bn=16, bk=32, wm=16, wn=16, mfma="f32_16x16x4"),
     medium=dict(bm=32, bn=32, bk=16, wm=32, wn=32, mfma="f32_32x32x2"),
-    large=dict(bm=64, bn=64, bk=16, wm=64, wn=64, mfma="f32_32x32x2"),
+    # large: 2 waves of 32x64 beat the single 64x64 wave by ~6% plain and
+    # ~11% fused at N=4096 (probe_pipeline PROBE_ONLY=T)
+    large=dict(bm=64, bn=64, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
     tall=dict(bm=128, bn=32, bk=16, wm=64, wn=32, mfma="f32_32x32x2"),
     wide=dict(bm=32, bn=128, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
     # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32

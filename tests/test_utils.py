@@ -42,9 +42,10 @@ def test_kernel_table_parity():
     assert KERNEL_NAMES[16] == "abft_kernel_huge"
     assert PERF_SWEEP_IDS == [0, 1, 2, 3, 4, 5, 6, 10, 11, 12, 13, 14, 15, 16]
     assert 7 not in KERNEL_TABLE and 9 not in KERNEL_TABLE
-    # wavefront-64 blocks, reference block-size parity
-    # (64/64/64/128/128/256 threads)
-    assert [threads(t) for t in TILING] == [64, 64, 64, 128, 128, 256]
+    # wavefront-64 blocks; sizes tuned for CDNA4 (large got a second wave,
+    # probe-measured +6% plain / +11% fused) — the reference's 32-thread
+    # warp sizing does not transfer
+    assert [threads(t) for t in TILING] == [64, 64, 128, 128, 128, 256]
     for t in TILING.values():
         assert t["bm"] % t["wm"] == 0 and t["bn"] % t["wn"] == 0
 

@@ -429,6 +429,17 @@ int main(int argc, char** argv) {
             "large   bk16 abft+inj OCC3", n, dA, dB, dC, dRef, dMax, reps);
         run_lib<32, 32, 16, 32, 32, true, true, false, false, -15, 32, 4>(
             "medium  bk16 abft+inj OCC4", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 64, 16, 32, 64, false, false, false, false, -15>(
+            "large   bk16 w2 WM32xWN64 plain", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<64, 64, 16, 32, 64, true, true, false, false, -15>(
+            "large   bk16 w2 WM32xWN64 abft+inj", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<256, 128, 16, 128, 64, true, true, false, false, -15>(
+            "huge    bk16 abft+inj (enc-asm)", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<256, 128, 16, 128, 64, false, false, false, false, -15>(
+            "huge    bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
         continue;
       }
       run_variant<32, 0, false>("P0 bk32 2buf syncthreads", n, dA, dB, dC,
