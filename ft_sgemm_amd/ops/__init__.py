@@ -72,6 +72,48 @@ def baseline_ft(a, b, c, alpha: float = 1.0, beta: float = 0.0,
     return c, res
 
 
+def choose_tier(m: int, n: int, k: int) -> str:
+    """Pick the fastest applicable tier for a problem shape.
+
+    Heuristic from the measured 1024..6144 sweep (profiles/overhead_table
+    sweeps): the huge 256x128 macro-tile wins once its grid fills the 512
+    co-resident block slots of the 256-CU chip reasonably evenly; at
+    straggler sizes (e.g. 3072: 288 blocks over 512 slots) the large 64x64
+    tier's fine grid wins; small shapes go to the small/medium tiers."""
+    from .. import kernel_table as kt
+
+    def fits(tier):
+        t = kt.TILING[tier]
+        return m % t["bm"] == 0 and n % t["bn"] == 0 and k % t["bk"] == 0
+
+    if fits("huge"):
+        blocks = (m // 256) * (n // 128)
+        rem = blocks % 512
+        if blocks >= 512 and (rem == 0 or rem >= 256 or blocks >= 2048):
+            return "huge"
+    for tier in ("tall" if m >= 4 * n else "wide" if n >= 4 * m else "large",
+                 "large", "medium", "small"):
+        if fits(tier):
+            return tier
+    raise ValueError(f"no tier fits M={m} N={n} K={k}")
+
+
+def sgemm_auto(a, b, c, alpha: float = 1.0, beta: float = 0.0):
+    """Plain SGEMM with automatic tier selection."""
+    k, m = a.shape
+    n = b.shape[1]
+    return sgemm(choose_tier(m, n, k), a, b, c, alpha, beta)
+
+
+def ft_sgemm_auto(a, b, c, alpha: float = 1.0, beta: float = 0.0,
+                  inject: bool = True):
+    """Fused-ABFT SGEMM with automatic tier selection."""
+    k, m = a.shape
+    n = b.shape[1]
+    return ft_sgemm(choose_tier(m, n, k), a, b, c, alpha, beta,
+                    inject=inject)
+
+
 def run_kernel_id(kid: int, a, b, c, alpha: float = 1.0, beta: float = 0.0,
                   inject: bool = True):
     """Dispatch by reference kernel id (0=rocBLAS, 1-6 plain tiers,

@@ -57,3 +57,22 @@ def test_perf_table_format():
     lines = text.splitlines()
     assert lines[0].startswith("Matrix Size|")
     assert "4695" in lines[1] and lines[1].endswith("|")
+
+
+def test_choose_tier():
+    from ft_sgemm_amd.ops import choose_tier
+    # big even shapes fill the huge grid
+    assert choose_tier(4096, 4096, 4096) == "huge"
+    assert choose_tier(8192, 8192, 8192) == "huge"
+    # straggler grid (3072: 288 blocks over 512 slots) -> large
+    assert choose_tier(3072, 3072, 3072) == "large"
+    # small/skinny shapes
+    assert choose_tier(1024, 1024, 1024) == "large"
+    assert choose_tier(512, 64, 256) == "tall"
+    assert choose_tier(64, 512, 256) == "wide"
+    assert choose_tier(64, 64, 64) == "large"
+    assert choose_tier(32, 32, 32) == "medium"
+    assert choose_tier(16, 16, 32) == "small"
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        choose_tier(17, 16, 32)
