@@ -135,3 +135,51 @@ def test_block_row_fused_abft_world1():
                     beta=0.0)
     torch.cuda.synchronize()
     check(ref, c)
+
+
+@pytest.mark.parametrize("k", [64, 128, 192, 320])
+def test_ft_small_k_strip_windows(k):
+    """ABFT strip-window edges: K smaller than / not aligned to the 64-k
+    strip load (PPS panels per window, odd window counts)."""
+    _require_native()
+    m, n = 256, 128
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_ft_high_fault_rate():
+    """64 verify windows -> 64 injected faults in one GEMM, all corrected
+    (well beyond the reference's 20-fault protocol)."""
+    _require_native()
+    m = n = k = 2048
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True, verify_windows=64)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_ft_custom_tau_and_magnitude():
+    """Threshold and injection magnitude are runtime parameters: a small
+    200.0 fault is corrected with tau=100 (reference hard-codes 9500/1e4)."""
+    _require_native()
+    m = n = k = 512
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True, tau=100.0,
+                 inj_mag=200.0)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_auto_tier_gpu():
+    _require_native()
+    for m, n, k in [(3072, 3072, 1024), (512, 512, 512), (64, 64, 64)]:
+        a, b, c = ops.make_operands(m, n, k)
+        ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+        ops.ft_sgemm_auto(a, b, c)
+        torch.cuda.synchronize()
+        check(ref, c)
