@@ -155,7 +155,8 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
   // saw(64)] window of the precomputed segment sums, streamed by
   // global_load_lds.  Only A-side sums are needed: the column checksums
   // detect, the weighted ones locate the row, and the column is lane-local.
-  static_assert(!(ABFT && PIPE), "3-buffer ring is plain-only");
+  static_assert(!(ABFT && PIPE) && !(INJECT && PIPE),
+                "3-buffer ring is plain-only");
   constexpr int NBUF = PIPE ? 3 : 2;
   constexpr int STRIP_OFF = NBUF * BUF;
   constexpr int LDS_FLOATS = ABFT ? (STRIP_OFF + NWAVES * 256)
