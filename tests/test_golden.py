@@ -128,3 +128,33 @@ def test_ratio_locate_clean_no_false_positive():
     corrected, locs = golden.abft_ratio_locate(prod, a, b, seg=64)
     assert locs == []
     assert np.array_equal(corrected, prod)
+
+
+def test_ratio_locate_property_random_faults():
+    """Property test: for random shapes, fault counts, sites and magnitudes
+    (above threshold), the ratio locate finds every site exactly and the
+    corrected product verifies against fp64."""
+    rng = np.random.default_rng(7)
+    for trial in range(25):
+        seg = int(rng.choice([16, 32, 64, 128]))
+        m = seg * int(rng.integers(1, 5))
+        n = int(rng.integers(1, 5)) * 32
+        k = int(rng.integers(2, 9)) * 32
+        a = generate_random_matrix(m, k, rng=rng)
+        b = generate_random_matrix(n, k, rng=rng)
+        prod = (a @ b.T).astype(np.float32)
+        # distinct (band, column) sites: one fault per band-column pair
+        nfaults = int(rng.integers(0, 4))
+        sites = {}
+        for _ in range(nfaults):
+            i, j = int(rng.integers(m)), int(rng.integers(n))
+            sites[(i // seg, j)] = (i, j)
+        mag = float(rng.uniform(1.2, 3.0)) * 1e4
+        for i, j in sites.values():
+            prod[i, j] += np.float32(mag * (1 if rng.random() < 0.5 else -1))
+        corrected, locs = golden.abft_ratio_locate(prod, a, b, seg=seg)
+        assert sorted(locs) == sorted(sites.values()), (
+            f"trial {trial}: {locs} vs {list(sites.values())}")
+        ref = a.astype(np.float64) @ b.astype(np.float64).T
+        ok, idx, _ = verify_matrix(ref, corrected)
+        assert ok, f"trial {trial}: mismatch at {idx}"
