@@ -70,14 +70,6 @@ template <> struct mfma_traits<16> {
   }
 };
 
-// Butterfly sum across the MM-lane sub-groups (sums a value over the lanes
-// that share one k-slice): masks 1..MM/2 stay inside the group.
-template <int MM> __device__ inline float group_sum(float v) {
-#pragma unroll
-  for (int m = 1; m < MM; m <<= 1) v += __shfl_xor(v, m, 64);
-  return v;
-}
-
 // Completes a per-k-slice partial into a full sum across the 64/MM slices
 // (masks MM..32).
 template <int MM> __device__ inline float slice_sum(float v) {
