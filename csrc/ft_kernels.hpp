@@ -78,6 +78,19 @@ template <int MM> __device__ inline float slice_sum(float v) {
   return v;
 }
 
+// Double-precision variant for the cold locate path (shuffles the two
+// 32-bit halves).
+template <int MM> __device__ inline double dslice_sum(double v) {
+#pragma unroll
+  for (int m = MM; m < 64; m <<= 1) {
+    union { double d; int i[2]; } u{v};
+    u.i[0] = __shfl_xor(u.i[0], m, 64);
+    u.i[1] = __shfl_xor(u.i[1], m, 64);
+    v += u.d;
+  }
+  return v;
+}
+
 // Accumulator register -> row within the MM x MM fragment (valid for both
 // MFMA shapes used here; for MM=16, reg>>2 == 0).
 __device__ constexpr int acc_row(int reg, int sub) {
@@ -270,17 +283,24 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
     asm volatile("" : "+v"(sub_o));
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
-      float colp = 0.f, colw = 0.f;
+      // fp64 column sums: the fault itself (|e| ~ 1e4) sits in this sum, so
+      // an fp32 accumulation rounds every later term at ulp(1e4) ~ 1e-3 and
+      // the CORRECTION inherits ~5e-3..1e-2 of error (found by
+      // tools/soak.py at |alpha| > 1).  Doubles are fine here: this is the
+      // cold path, entered only by a wave that absorbed a fault.
+      double colp = 0.0, colw = 0.0;
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
         for (int reg = 0; reg < NREG; ++reg) {
-          const float v = acc[fm][fn][reg];
+          const double v = (double)acc[fm][fn][reg];
           colp += v;
-          colw = fmaf((float)(fm * MM + acc_row(reg, sub_o)), v, colw);
+          colw = fma((double)(fm * MM + acc_row(reg, sub_o)), v, colw);
         }
-      const float rc = slice_sum<MM>(colp) - slice_sum<MM>(cc[fn]);
-      const float rw = slice_sum<MM>(colw) - slice_sum<MM>(cw[fn]);
+      const float rc =
+          (float)(dslice_sum<MM>(colp) - (double)slice_sum<MM>(cc[fn]));
+      const float rw =
+          (float)(dslice_sum<MM>(colw) - (double)slice_sum<MM>(cw[fn]));
       const bool cbad = fabsf(rc) > tau;
       const int row = (int)rintf(rw / (cbad ? rc : 1.f));
 #pragma unroll
