@@ -38,17 +38,37 @@ def main():
         inject = rng.random() < 0.7
         vw = rng.choice([1, 5, 20, 40])
         seed = rng.randint(0, 10_000)
+        op = rng.choice(["ft", "ft", "ft", "plain", "baseline", "blockrow",
+                         "auto"])
         a, b, c = ops.make_operands(m, n, k, seed=seed)
         if beta != 0.0:
             c.normal_(0, 1.0)
         ref = ops.torch_reference(a, b, c, alpha, beta)
-        ops.ft_sgemm(tier, a, b, c, alpha, beta, inject=inject,
-                     verify_windows=vw)
+        if op == "ft":
+            ops.ft_sgemm(tier, a, b, c, alpha, beta, inject=inject,
+                         verify_windows=vw)
+        elif op == "plain":
+            ops.sgemm(tier, a, b, c, alpha, beta)
+        elif op == "baseline":
+            pk = rng.choice([256, 512, 1024])
+            ops.baseline_ft(a, b, c, alpha, beta, panel_k=pk)
+        elif op == "blockrow":
+            from ft_sgemm_amd.parallel import block_row_sgemm
+            pk = k if k % 64 else k  # single rank: any panel dividing k
+            for cand in (64, 128, 256):
+                if k % cand == 0:
+                    pk = cand
+            block_row_sgemm(
+                a, b, c, panel_k=pk, alpha=alpha, beta=beta,
+                gemm_fn=lambda ap, bp, cl, al, be: ops.ft_sgemm(
+                    tier, ap, bp, cl, al, be, inject=inject))
+        else:
+            ops.ft_sgemm_auto(a, b, c, alpha, beta, inject=inject)
         torch.cuda.synchronize()
         diff = (ref - c).abs()
         rel = diff / ref.abs().clamp_min(1e-30)
         bad = int(((diff > 1e-2) & (rel > 1e-2)).sum())
-        line = (f"trial {t}: tier={tier} m={m} n={n} k={k} alpha={alpha} "
+        line = (f"trial {t}: op={op} tier={tier} m={m} n={n} k={k} alpha={alpha} "
                 f"beta={beta} inject={inject} vw={vw} seed={seed} "
                 f"bad={bad} maxdiff={diff.max().item():.2e}")
         if bad:
