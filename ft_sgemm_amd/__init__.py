@@ -6,10 +6,11 @@ shixun404/Fault-Tolerant-SGEMM-on-NVIDIA-GPUs (arXiv:2305.01024):
 * six hand-tiled fp32 GEMM kernels (small/medium/large/tall/wide/huge)
   built on f32-input MFMA (`v_mfma_f32_32x32x2_f32` / `v_mfma_f32_16x16x4_f32`)
   with LDS double-buffered A/B panels,
-* fused-ABFT twins that maintain wave-level row/column checksums of the
-  output tile in registers, periodically verify them with cross-lane
-  reductions, locate a corrupted accumulator element at the row x column
-  residual intersection and correct it in place,
+* fused-ABFT twins that maintain wave-level column checksums (plain +
+  row-index-weighted) of the output tile in registers from precomputed
+  segment sums, periodically run a cheap whole-tile detect, locate a
+  corrupted accumulator element by the residual ratio (column is
+  lane-local, row = round(rw/rc)) and correct it in place,
 * a non-fused ABFT baseline composed from rocBLAS calls,
 * a rocBLAS oracle path (kernel id 0),
 * multi-GPU scaling via torch.distributed over RCCL/xGMI
