@@ -260,7 +260,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
 
   // ---- ABFT verify / locate / correct: wave-autonomous, registers only ----
   // Two-phase: a cheap detect (compare the total tile sum against the total
-  // checksum — ~70 VALU ops) runs every verify window; the locate/correct
+  // checksum — FM*FN*NREG adds + a butterfly) runs every verify window; the locate/correct
   // below is entered only when the residual trips the threshold (i.e. in
   // the wave that actually absorbed a fault), so the fault-free common case
   // never pays for location.
@@ -314,7 +314,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
   };
 
   auto verify_correct = [&]() __attribute__((always_inline)) {
-    // Cheap detect: total tile sum vs total checksum (~70 VALU + 6
+    // Cheap detect: total tile sum vs total checksum (FM*FN*NREG adds + 6
     // shuffles).  The residual is identical across the wave's lanes, so
     // the branch is uniform; only a wave that actually absorbed a fault
     // enters the locate/correct path above.
@@ -335,11 +335,10 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
 
   // ---- main K loop: one barrier per BK panel, glds prefetch overlaps ----
   // Structure: bursts of `verify_iters` panels.  The inner panel loop never
-  // touches the accumulator with VALU code, so the compiler keeps it in
-  // AGPRs with zero v_accvgpr traffic; injection and verify/correct run at
-  // burst boundaries only (a conditional VALU read of acc inside the panel
-  // loop makes hipcc shuttle all 64 accumulator registers AGPR<->VGPR every
-  // panel — measured 2x wall time on the huge tier).
+  // touches the accumulator with VALU code; injection and verify/correct
+  // run at burst boundaries only (a conditional VALU read of acc inside the
+  // panel loop makes hipcc shuttle the whole accumulator file through VGPR
+  // copies every panel — measured 2x wall time on the huge tier).
   // A strip load covers 64 k values = PPS panels; strips are double
   // buffered by strip-window parity.
   constexpr int PPS = 64 / BK;  // panels per strip window (BK <= 64)
