@@ -440,6 +440,16 @@ int main(int argc, char** argv) {
             reps);
         run_lib<256, 128, 16, 128, 64, false, false, false, false, -15>(
             "huge    bk16 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 64, 16, 64, 64, false, false, false, false, -15>(
+            "tall2   128x64x16 w2 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<128, 64, 16, 64, 64, true, true, false, false, -15>(
+            "tall2   128x64x16 w2 abft+inj", n, dA, dB, dC, dRef, dMax,
+            reps);
+        run_lib<64, 128, 16, 64, 64, false, false, false, false, -15>(
+            "wide2   64x128x16 w2 plain", n, dA, dB, dC, dRef, dMax, reps);
+        run_lib<64, 128, 16, 64, 64, true, true, false, false, -15>(
+            "wide2   64x128x16 w2 abft+inj", n, dA, dB, dC, dRef, dMax,
+            reps);
         continue;
       }
       run_variant<32, 0, false>("P0 bk32 2buf syncthreads", n, dA, dB, dC,
