@@ -94,9 +94,19 @@ def main():
     else:
         step = lambda: ops.rocblas_sgemm(a, b, c, 1.0, -1.5)
 
+    warm0 = time.perf_counter()
     for _ in range(args.warmup):
         step()
     if not cpu_ci:
+        torch.cuda.synchronize()
+    # clock-ramp extension: a cold MI355X takes ~1-2 s to reach its steady
+    # compute clock; short warmups would time the ramp.  Extra untimed steps
+    # (beyond the requested W) are reported in config.warmup_extra.
+    warmup_extra = 0
+    if not cpu_ci:
+        while time.perf_counter() - warm0 < 1.5 and warmup_extra < 2000:
+            step()
+            warmup_extra += 1
         torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
@@ -144,6 +154,7 @@ def main():
                 "M": n, "N": n, "K": n,
                 "alpha": 1.0, "beta": -1.5,
                 "inject": True, "faults_per_gemm": 20,
+                "warmup_extra": warmup_extra,
                 "parallelism": (f"dp{world}" if args.mode == "replicated"
                                 else f"blockrow{world}"),
             },
