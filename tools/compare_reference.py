@@ -36,7 +36,7 @@ REF_4096 = _ref_table()
 
 def main():
     path = sys.argv[1] if len(sys.argv) > 1 else os.path.join(
-        ROOT, "profiles", "cli_sweep_final.log")
+        ROOT, "profiles", "cli_sweep_final2.log")
     rows = {}
     sizes = None
     for line in open(path):
