@@ -158,3 +158,29 @@ def test_ratio_locate_property_random_faults():
         ref = a.astype(np.float64) @ b.astype(np.float64).T
         ok, idx, _ = verify_matrix(ref, corrected)
         assert ok, f"trial {trial}: mismatch at {idx}"
+
+
+def test_repeated_same_site_correction_residue_bound():
+    """Each ABFT correction leaves O(checksum roundoff) residue at the
+    corrected element; repeated faults at the SAME site across panel-GEMMs
+    accumulate it linearly.  Golden model: the residue stays ~1e-4/panel
+    (the device path matches after the fp64 cold-path fix)."""
+    rng = np.random.default_rng(5)
+    m = n = 64
+    total = np.zeros((m, n), dtype=np.float32)
+    exact = np.zeros((m, n), dtype=np.float64)
+    panels = 32
+    for p in range(panels):
+        a = generate_random_matrix(m, 64, rng=rng)
+        b = generate_random_matrix(n, 64, rng=rng)
+        prod = (a @ b.T).astype(np.float32)
+        prod[0, 0] += np.float32(1e4)       # same site every panel
+        corrected, locs = golden.abft_ratio_locate(prod, a, b, seg=64)
+        assert locs == [(0, 0)]
+        total += corrected
+        exact += a.astype(np.float64) @ b.astype(np.float64).T
+    residue = abs(float(total[0, 0]) - float(exact[0, 0]))
+    # linear-in-panels accumulation, ~1e-4 per correction; assert the bound
+    assert residue < panels * 2e-3
+    ok, idx, _ = verify_matrix(exact, total)
+    assert ok, f"accumulated residue broke tolerance at {idx}"
