@@ -205,6 +205,325 @@ __global__ __launch_bounds__(256, 2) void k_dtva2(int M, int N, int K,
     }
 }
 
+
+
+// k_dtvasm: the Tensile-structure SGEMM with ALL VMEM in inline asm so
+// hipcc's conservative per-element vmcnt tracking (36 near-zero waits per
+// 64 MFMAs in k_dtva2, measured) is replaced by two hand-counted waits per
+// panel.  BM=256 BN=128, 4 waves of 64x128 (FM=2,FN=4); A slices are
+// wave-private and go straight to registers; only B transits LDS.
+// Guide form (ii): "=&v" loads in one statement, then a wait statement
+// naming every destination "+v" before the first consumer (s_nop 1 for the
+// VALU->MFMA boundary hazard).  BKT in {8, 16}.
+template <int BKT>
+__global__ __launch_bounds__(256, 2) void k_dtvasm(
+    int M, int N, int K, const float* __restrict__ A,
+    const float* __restrict__ B, float* __restrict__ C, float alpha,
+    float beta) {
+  __shared__ __attribute__((aligned(16))) float Bs[2 * 128 * BKT];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int sub = lane >> 5, r = lane & 31;
+  const int wi0 = wave * 64;
+  const int im0 = blockIdx.x * 256, jn0 = blockIdx.y * 128;
+
+  constexpr int NA = BKT;          // A dwords per panel per lane (2fm*BKT/2)
+  constexpr int NB = BKT / 8;      // B dwordx4 per panel per thread
+  constexpr int NV = NA + NB;      // VMEM ops per panel
+
+  f32x16 acc[2][4] = {};
+  float aA[NA], aB[NA];
+  f32x4 breg[NB];
+
+  const int fB = tid * 4;          // B chunk 0; chunk 1 (BKT=16) at +1024
+  const float* pB0 = B + (jn0 + (fB & 127)) + (size_t)(fB >> 7) * N;
+  const float* pA0 = A + (im0 + wi0 + r) + (size_t)sub * M;
+
+  // A fragment (fm,kk) register index: ao[fm*4+kk] per 8-k half; the
+  // offset:128 forms load fm=1 (row +32 floats)
+
+#define ISSUE8H(pb, q0, q1, q2, q3, b0v, aoff)                             \
+  asm volatile("global_load_dwordx4 %0, %9, off\n\t"                        \
+               "global_load_dword %1, %10, off\n\t"                         \
+               "global_load_dword %2, %11, off\n\t"                         \
+               "global_load_dword %3, %12, off\n\t"                         \
+               "global_load_dword %4, %13, off\n\t"                         \
+               "global_load_dword %5, %10, off offset:128\n\t"              \
+               "global_load_dword %6, %11, off offset:128\n\t"              \
+               "global_load_dword %7, %12, off offset:128\n\t"              \
+               "global_load_dword %8, %13, off offset:128"                  \
+               : "=&v"(b0v), "=&v"((aoff)[0]), "=&v"((aoff)[1]),            \
+                 "=&v"((aoff)[2]), "=&v"((aoff)[3]), "=&v"((aoff)[4]),      \
+                 "=&v"((aoff)[5]), "=&v"((aoff)[6]), "=&v"((aoff)[7])       \
+               : "v"(pb), "v"(q0), "v"(q1), "v"(q2), "v"(q3))
+
+#define ISSUE(p, bo, ao)                                                    \
+  do {                                                                      \
+    const size_t ko = (size_t)(p) * BKT;                                    \
+    const float* pb = pB0 + ko * N;                                         \
+    const float* q0 = pA0 + ko * M;                                         \
+    const float* q1 = q0 + 2 * (size_t)M;                                   \
+    const float* q2 = q0 + 4 * (size_t)M;                                   \
+    const float* q3 = q0 + 6 * (size_t)M;                                   \
+    ISSUE8H(pb, q0, q1, q2, q3, (bo)[0], ao);                               \
+    if constexpr (BKT == 16) {                                              \
+      const float* pb2 = pb + 8 * (size_t)N;                                \
+      const float* q4 = q0 + 8 * (size_t)M;                                 \
+      const float* q5 = q0 + 10 * (size_t)M;                                \
+      const float* q6 = q0 + 12 * (size_t)M;                                \
+      const float* q7 = q0 + 14 * (size_t)M;                                \
+      ISSUE8H(pb2, q4, q5, q6, q7, (bo)[NB - 1], (ao) + 8);                 \
+    }                                                                       \
+  } while (0)
+
+#define WAIT_A(n, ao)                                                       \
+  do {                                                                      \
+    asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1"                       \
+                 : "+v"((ao)[0]), "+v"((ao)[1]), "+v"((ao)[2]),             \
+                   "+v"((ao)[3]), "+v"((ao)[4]), "+v"((ao)[5]),             \
+                   "+v"((ao)[6]), "+v"((ao)[7]));                           \
+    if constexpr (BKT == 16)                                                \
+      asm volatile("" : "+v"((ao)[8]), "+v"((ao)[9]), "+v"((ao)[10]),       \
+                     "+v"((ao)[11]), "+v"((ao)[12]), "+v"((ao)[13]),        \
+                     "+v"((ao)[14]), "+v"((ao)[15]));                       \
+  } while (0)
+#define WAIT_B(n, bo)                                                       \
+  do {                                                                      \
+    asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1" : "+v"((bo)[0]));     \
+    if constexpr (BKT == 16) asm volatile("" : "+v"((bo)[1]));              \
+  } while (0)
+
+  const int niter = K / BKT;
+  ISSUE(0, breg, aA);
+  WAIT_A(0, aA);  // prologue: drain everything (breg implied)
+  WAIT_B(0, breg);
+  *(f32x4*)(&Bs[0] + fB) = breg[0];
+  if constexpr (BKT == 16) *(f32x4*)(&Bs[0] + fB + 1024) = breg[1];
+  __syncthreads();
+
+#define KKLOOP(q, ao)                                                       \
+  do {                                                                      \
+    const float* Bp = &Bs[(q) * 128 * BKT];                                 \
+    _Pragma("unroll") for (int kk = 0; kk < BKT / 2; ++kk) {                \
+      const int kloc = kk * 2 + sub;                                        \
+      float b[4];                                                           \
+      _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                      \
+          b[fn] = Bp[kloc * 128 + fn * 32 + r];                             \
+      _Pragma("unroll") for (int fm = 0; fm < 2; ++fm)                      \
+          _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                  \
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_32x32x2f32(           \
+                  (ao)[(kk / 4) * 8 + fm * 4 + (kk % 4)], b[fn],            \
+                  acc[fm][fn], 0, 0, 0);                                    \
+    }                                                                       \
+  } while (0)
+
+#define BODY(it, acur, anxt)                                                \
+  do {                                                                      \
+    const int q = (it) & 1;                                                 \
+    if ((it) + 1 < niter) {                                                 \
+      ISSUE((it) + 1, breg, anxt);                                          \
+      if constexpr (BKT == 8) WAIT_A(9, acur);                              \
+      else WAIT_A(18, acur);                                                \
+    } else {                                                                \
+      WAIT_A(0, acur);                                                      \
+    }                                                                       \
+    KKLOOP(q, acur);                                                        \
+    if ((it) + 1 < niter) {                                                 \
+      if constexpr (BKT == 8) {                                             \
+        WAIT_B(8, breg);                                                    \
+        *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = breg[0];                 \
+      } else {                                                              \
+        WAIT_B(8, breg); /* retires B0(pos1) and B1(pos10) of 18 */         \
+        *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = breg[0];                 \
+        *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB + 1024) = breg[1];          \
+      }                                                                     \
+    }                                                                       \
+    __syncthreads();                                                        \
+  } while (0)
+
+  for (int it = 0; it < niter; it += 2) {
+    BODY(it, aA, aB);
+    BODY(it + 1, aB, aA);
+  }
+#undef ISSUE8
+#undef ISSUE
+#undef WAIT_A
+#undef WAIT_B
+#undef KKLOOP
+#undef BODY
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      const int j = jn0 + fn * 32 + r;
+      float* colbase = C + (size_t)j * M + im0 + wi0 + fm * 32;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        float* p = colbase + 4 * sub + 8 * g;
+        const f32x4 prev = *(const f32x4*)p;
+        f32x4 out;
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          out[u] = alpha * acc[fm][fn][4 * g + u] + beta * prev[u];
+        *(f32x4*)p = out;
+      }
+    }
+}
+
+
+// k_dtvasm2: depth-2 VMEM pipeline — panels it+1 and it+2 stay in flight
+// across the barrier; ONE hand-counted vmcnt(8) per panel covers both the
+// B publish and the A consume (waits are "<=N outstanding", so the count
+// is entry-state independent).  A register sets rotate with period 3,
+// B staging regs with period 2 -> 6-body unroll with guarded tails.
+__global__ __launch_bounds__(256, 1) void k_dtvasm2(
+    int M, int N, int K, const float* __restrict__ A,
+    const float* __restrict__ B, float* __restrict__ C, float alpha,
+    float beta) {
+  constexpr int BKT = 8;
+  __shared__ __attribute__((aligned(16))) float Bs[2 * 128 * BKT];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int sub = lane >> 5, r = lane & 31;
+  const int wi0 = wave * 64;
+  const int im0 = blockIdx.x * 256, jn0 = blockIdx.y * 128;
+
+  f32x16 acc[2][4] = {};
+  float aA[8], aB[8], aC[8];
+  f32x4 br0, br1;  // panel p's staged B quarter lives in br[p&1]
+
+  const int fB = tid * 4;
+  // SRSRC buffer addressing (T8): one descriptor per operand in SGPRs,
+  // per-lane 32-bit voffset, per-panel k-advance through the SGPR soffset.
+  const auto rsA = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)A, (short)0, 0xffffffffu, 0x00020000);
+  const auto rsB = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)B, (short)0, 0xffffffffu, 0x00020000);
+  const unsigned voffA = 4u * (unsigned)(im0 + wi0 + r + sub * (size_t)M);
+  const unsigned voffB =
+      4u * (unsigned)(jn0 + (fB & 127) + (size_t)(fB >> 7) * N);
+  const unsigned sM8 = 8u * (unsigned)M;   // 2*M floats in bytes
+  const unsigned sN32 = 32u * (unsigned)N; // BKT*N floats in bytes
+
+#define ISSUE(p, b0v, ao)                                                   \
+  do {                                                                      \
+    const unsigned kB = (unsigned)(p) * sN32;                               \
+    const unsigned k0 = (unsigned)(p) * BKT * 4u * (unsigned)M;             \
+    const unsigned k1 = k0 + sM8;                                           \
+    const unsigned k2 = k0 + 2 * sM8;                                       \
+    const unsigned k3 = k0 + 3 * sM8;                                       \
+    asm volatile("buffer_load_dwordx4 %0, %9, %10, %11 offen\n\t"          \
+                 "buffer_load_dword %1, %12, %13, %14 offen\n\t"           \
+                 "buffer_load_dword %2, %12, %13, %15 offen\n\t"           \
+                 "buffer_load_dword %3, %12, %13, %16 offen\n\t"           \
+                 "buffer_load_dword %4, %12, %13, %17 offen\n\t"           \
+                 "buffer_load_dword %5, %12, %13, %14 offen offset:128\n\t" \
+                 "buffer_load_dword %6, %12, %13, %15 offen offset:128\n\t" \
+                 "buffer_load_dword %7, %12, %13, %16 offen offset:128\n\t" \
+                 "buffer_load_dword %8, %12, %13, %17 offen offset:128"     \
+                 : "=&v"(b0v), "=&v"((ao)[0]), "=&v"((ao)[1]),              \
+                   "=&v"((ao)[2]), "=&v"((ao)[3]), "=&v"((ao)[4]),          \
+                   "=&v"((ao)[5]), "=&v"((ao)[6]), "=&v"((ao)[7])           \
+                 : "v"(voffB), "s"(rsB), "s"(kB), "v"(voffA), "s"(rsA),     \
+                   "s"(k0), "s"(k1), "s"(k2), "s"(k3));                     \
+  } while (0)
+#define WAIT_AB(n, ao, bv)                                                  \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1"                         \
+               : "+v"((ao)[0]), "+v"((ao)[1]), "+v"((ao)[2]),               \
+                 "+v"((ao)[3]), "+v"((ao)[4]), "+v"((ao)[5]),               \
+                 "+v"((ao)[6]), "+v"((ao)[7]), "+v"(bv))
+#define WAIT_A(n, ao)                                                       \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1"                         \
+               : "+v"((ao)[0]), "+v"((ao)[1]), "+v"((ao)[2]),               \
+                 "+v"((ao)[3]), "+v"((ao)[4]), "+v"((ao)[5]),               \
+                 "+v"((ao)[6]), "+v"((ao)[7]))
+#define WAIT_B(n, bv) \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1" : "+v"(bv))
+
+#define KKLOOP(q, ao)                                                       \
+  do {                                                                      \
+    const float* Bp = &Bs[(q) * 128 * BKT];                                 \
+    _Pragma("unroll") for (int kk = 0; kk < BKT / 2; ++kk) {                \
+      const int kloc = kk * 2 + sub;                                        \
+      float b[4];                                                           \
+      _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                      \
+          b[fn] = Bp[kloc * 128 + fn * 32 + r];                             \
+      _Pragma("unroll") for (int fm = 0; fm < 2; ++fm)                      \
+          _Pragma("unroll") for (int fn = 0; fn < 4; ++fn)                  \
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_32x32x2f32(           \
+                  (ao)[fm * 4 + kk], b[fn], acc[fm][fn], 0, 0, 0);          \
+    }                                                                       \
+  } while (0)
+
+  // body it: consume Acur + Bs[it&1]; publish BRwr (panel it+1's B) to
+  // Bs[(it+1)&1]; issue panel it+2 into (BRiss, Anxt).
+#define BODY(it, Acur, Anxt, BRwr, BRiss)                                   \
+  do {                                                                      \
+    const int q = (it) & 1;                                                 \
+    if ((it) + 2 < niter) {                                                 \
+      WAIT_AB(8, Acur, BRwr);                                               \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRwr;                      \
+      ISSUE((it) + 2, BRiss, Anxt);                                         \
+    } else if ((it) + 1 < niter) {                                          \
+      WAIT_AB(8, Acur, BRwr);                                               \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRwr;                      \
+    } else {                                                                \
+      WAIT_A(0, Acur);                                                      \
+    }                                                                       \
+    KKLOOP(q, Acur);                                                        \
+    __syncthreads();                                                        \
+  } while (0)
+
+  const int niter = K / BKT;  // caller guarantees niter >= 3
+  ISSUE(0, br0, aA);
+  ISSUE(1, br1, aB);
+  WAIT_B(17, br0);
+  *(f32x4*)(&Bs[0] + fB) = br0;
+  __syncthreads();
+
+  int it = 0;
+  while (it < niter) {
+    BODY(it, aA, aC, br1, br0);
+    if (++it >= niter) break;
+    BODY(it, aB, aA, br0, br1);
+    if (++it >= niter) break;
+    BODY(it, aC, aB, br1, br0);
+    if (++it >= niter) break;
+    BODY(it, aA, aC, br0, br1);
+    if (++it >= niter) break;
+    BODY(it, aB, aA, br1, br0);
+    if (++it >= niter) break;
+    BODY(it, aC, aB, br0, br1);
+    ++it;
+  }
+#undef ISSUE
+#undef WAIT_AB
+#undef WAIT_A
+#undef WAIT_B
+#undef KKLOOP
+#undef BODY
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      const int j = jn0 + fn * 32 + r;
+      float* colbase = C + (size_t)j * M + im0 + wi0 + fm * 32;
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        float* p = colbase + 4 * sub + 8 * g;
+        const f32x4 prev = *(const f32x4*)p;
+        f32x4 out;
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          out[u] = alpha * acc[fm][fn][4 * g + u] + beta * prev[u];
+        *(f32x4*)p = out;
+      }
+    }
+}
+
+
 __global__ void fill_lcg(float* p, size_t n, unsigned seed) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
@@ -307,6 +626,96 @@ int main(int argc, char** argv) {
               reps);
     run<32, 2>("D8 dtva2 256x128x32 1-barrier", n, dA, dB, dC, dRef, dMax,
                reps);
+    {  // asm-pipelined variant
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm<8>, grid, block, 0, 0, n, n, n, dA, dB, dC,
+                         1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm<8>, grid, block, 0, 0, n, n, n, dA, dB, dC,
+                           1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm<8>, grid, block, 0, 0, n, n, n, dA, dB, dC,
+                           1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D9 dtvasm 256x128x8 asm-vmem",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // depth-2 asm pipeline
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm2, grid, block, 0, 0, n, n, n, dA, dB, dC,
+                         1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm2, grid, block, 0, 0, n, n, n, dA, dB,
+                           dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm2, grid, block, 0, 0, n, n, n, dA, dB,
+                           dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D10 dtvasm2 depth-2 pipeline",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // BK=16 asm pipeline
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm<16>, grid, block, 0, 0, n, n, n, dA, dB,
+                         dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm<16>, grid, block, 0, 0, n, n, n, dA, dB,
+                           dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm<16>, grid, block, 0, 0, n, n, n, dA, dB,
+                           dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D11 dtvasm 256x128x16 asm-vmem",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
   }
   return 0;
 }
