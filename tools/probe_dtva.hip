@@ -524,6 +524,123 @@ __global__ __launch_bounds__(256, 1) void k_dtvasm2(
 }
 
 
+
+// k_dtvasm16: the D9 asm skeleton with rocBLAS's MFMA choice
+// (v_mfma_f32_16x16x4_f32): fragment rows land on 16-lane groups, so the
+// four fm offsets become immediate `offset:` fields (3 pointer inputs).
+__global__ __launch_bounds__(256, 2) void k_dtvasm16(
+    int M, int N, int K, const float* __restrict__ A,
+    const float* __restrict__ B, float* __restrict__ C, float alpha,
+    float beta) {
+  constexpr int BKT = 8;
+  __shared__ __attribute__((aligned(16))) float Bs[2 * 128 * BKT];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int s16 = lane >> 4, r16 = lane & 15;
+  const int wi0 = wave * 64;
+  const int im0 = blockIdx.x * 256, jn0 = blockIdx.y * 128;
+
+  f32x4 acc[4][8] = {};
+  float aA[8], aB[8];
+  f32x4 breg;
+
+  const int fB = tid * 4;
+  const float* pB0 = B + (jn0 + (fB & 127)) + (size_t)(fB >> 7) * N;
+  const float* pA0 = A + (im0 + wi0 + r16) + (size_t)s16 * M;
+
+#define ISSUE(p, b0v, ao)                                                   \
+  do {                                                                      \
+    const size_t ko = (size_t)(p) * BKT;                                    \
+    const float* pb = pB0 + ko * N;                                         \
+    const float* q0 = pA0 + ko * M;                                         \
+    const float* q1 = q0 + 4 * (size_t)M;                                   \
+    asm volatile("global_load_dwordx4 %0, %9, off\n\t"                      \
+                 "global_load_dword %1, %10, off\n\t"                       \
+                 "global_load_dword %2, %10, off offset:64\n\t"             \
+                 "global_load_dword %3, %10, off offset:128\n\t"            \
+                 "global_load_dword %4, %10, off offset:192\n\t"            \
+                 "global_load_dword %5, %11, off\n\t"                       \
+                 "global_load_dword %6, %11, off offset:64\n\t"             \
+                 "global_load_dword %7, %11, off offset:128\n\t"            \
+                 "global_load_dword %8, %11, off offset:192"                \
+                 : "=&v"(b0v), "=&v"((ao)[0]), "=&v"((ao)[1]),              \
+                   "=&v"((ao)[2]), "=&v"((ao)[3]), "=&v"((ao)[4]),          \
+                   "=&v"((ao)[5]), "=&v"((ao)[6]), "=&v"((ao)[7])           \
+                 : "v"(pb), "v"(q0), "v"(q1));                              \
+  } while (0)
+#define WAIT_A(n, ao)                                                       \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1"                         \
+               : "+v"((ao)[0]), "+v"((ao)[1]), "+v"((ao)[2]),               \
+                 "+v"((ao)[3]), "+v"((ao)[4]), "+v"((ao)[5]),               \
+                 "+v"((ao)[6]), "+v"((ao)[7]))
+#define WAIT_B(n) \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1" : "+v"(breg))
+
+  const int niter = K / BKT;
+  ISSUE(0, breg, aA);
+  WAIT_B(8);
+  WAIT_A(0, aA);
+  *(f32x4*)(&Bs[0] + fB) = breg;
+  __syncthreads();
+
+#define KKLOOP(q, ao)                                                       \
+  do {                                                                      \
+    const float* Bp = &Bs[(q) * 128 * BKT];                                 \
+    _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                      \
+      const int kloc = kk * 4 + s16;                                        \
+      float b[8];                                                           \
+      _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                      \
+          b[fn] = Bp[kloc * 128 + fn * 16 + r16];                           \
+      _Pragma("unroll") for (int fm = 0; fm < 4; ++fm)                      \
+          _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                  \
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x4f32(           \
+                  (ao)[kk * 4 + fm], b[fn], acc[fm][fn], 0, 0, 0);          \
+    }                                                                       \
+  } while (0)
+
+#define BODY(it, acur, anxt)                                                \
+  do {                                                                      \
+    const int q = (it) & 1;                                                 \
+    if ((it) + 1 < niter) {                                                 \
+      ISSUE((it) + 1, breg, anxt);                                          \
+      WAIT_A(9, acur);                                                      \
+    } else {                                                                \
+      WAIT_A(0, acur);                                                      \
+    }                                                                       \
+    KKLOOP(q, acur);                                                        \
+    if ((it) + 1 < niter) {                                                 \
+      WAIT_B(8);                                                            \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = breg;                      \
+    }                                                                       \
+    __syncthreads();                                                        \
+  } while (0)
+
+  for (int it = 0; it < niter; it += 2) {
+    BODY(it, aA, aB);
+    BODY(it + 1, aB, aA);
+  }
+#undef ISSUE
+#undef WAIT_A
+#undef WAIT_B
+#undef KKLOOP
+#undef BODY
+
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 8; ++fn) {
+      const int j = jn0 + fn * 16 + r16;
+      float* p = C + (size_t)j * M + im0 + wi0 + fm * 16 + s16 * 4;
+      const f32x4 prev = *(const f32x4*)p;
+      f32x4 out;
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        out[u] = alpha * acc[fm][fn][u] + beta * prev[u];
+      *(f32x4*)p = out;
+    }
+}
+
+
 __global__ void fill_lcg(float* p, size_t n, unsigned seed) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
@@ -713,6 +830,36 @@ int main(int argc, char** argv) {
       hipEventElapsedTime(&ms, b0, b1);
       printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
              "D11 dtvasm 256x128x16 asm-vmem",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // 16x16x4 MFMA on the asm skeleton
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm16, grid, block, 0, 0, n, n, n, dA, dB, dC,
+                         1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm16, grid, block, 0, 0, n, n, n, dA, dB,
+                           dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm16, grid, block, 0, 0, n, n, n, dA, dB,
+                           dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D12 dtvasm16 16x16x4 asm",
              2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
              hipGetErrorString(hipGetLastError()));
     }
