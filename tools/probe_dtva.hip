@@ -528,7 +528,8 @@ __global__ __launch_bounds__(256, 1) void k_dtvasm2(
 // k_dtvasm16: the D9 asm skeleton with rocBLAS's MFMA choice
 // (v_mfma_f32_16x16x4_f32): fragment rows land on 16-lane groups, so the
 // four fm offsets become immediate `offset:` fields (3 pointer inputs).
-__global__ __launch_bounds__(256, 2) void k_dtvasm16(
+template <int OCC = 2>
+__global__ __launch_bounds__(256, OCC) void k_dtvasm16(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
     float beta) {
@@ -836,7 +837,7 @@ int main(int argc, char** argv) {
     {  // 16x16x4 MFMA on the asm skeleton
       dim3 grid(n / BM, n / BN), block(256);
       hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
-      hipLaunchKernelGGL(k_dtvasm16, grid, block, 0, 0, n, n, n, dA, dB, dC,
+      hipLaunchKernelGGL(k_dtvasm16<2>, grid, block, 0, 0, n, n, n, dA, dB, dC,
                          1.f, 0.f);
       float md = 0.f;
       hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
@@ -847,12 +848,12 @@ int main(int argc, char** argv) {
       hipEventCreate(&b0);
       hipEventCreate(&b1);
       for (int w = 0; w < 2; ++w)
-        hipLaunchKernelGGL(k_dtvasm16, grid, block, 0, 0, n, n, n, dA, dB,
+        hipLaunchKernelGGL(k_dtvasm16<2>, grid, block, 0, 0, n, n, n, dA, dB,
                            dC, 1.f, -1.5f);
       hipDeviceSynchronize();
       hipEventRecord(b0);
       for (int rr = 0; rr < reps; ++rr)
-        hipLaunchKernelGGL(k_dtvasm16, grid, block, 0, 0, n, n, n, dA, dB,
+        hipLaunchKernelGGL(k_dtvasm16<2>, grid, block, 0, 0, n, n, n, dA, dB,
                            dC, 1.f, -1.5f);
       hipEventRecord(b1);
       hipEventSynchronize(b1);
@@ -860,6 +861,36 @@ int main(int argc, char** argv) {
       hipEventElapsedTime(&ms, b0, b1);
       printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
              "D12 dtvasm16 16x16x4 asm",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // OCC=3 variant (3 blocks/CU if it fits 168 VGPRs)
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm16<3>, grid, block, 0, 0, n, n, n, dA, dB,
+                         dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm16<3>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm16<3>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D13 dtvasm16 OCC3",
              2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
              hipGetErrorString(hipGetLastError()));
     }
