@@ -272,6 +272,18 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
   // intersection (ft_sgemm_huge.cuh:422-485) whose row residuals needed a
   // full cross-lane butterfly per accumulator register — the measured
   // window cost dropped ~10x because no row reductions exist at all.
+  //
+  // Guarantee (same as the reference): ONE fault per verify window per
+  // wave tile.  TWO faults e1 at row r1 and e2 at row r2 landing in the
+  // SAME column of one wave tile within one window degrade differently
+  // from the reference: rc = e1 + e2, row = round((r1*e1 + r2*e2) /
+  // (e1+e2)), and rc is subtracted IF some accumulator row matches that
+  // blended index — correcting a clean element by the combined magnitude
+  // (the reference's intersection scheme instead corrects every
+  // row-candidate x col-candidate).  When the blend rounds outside the
+  // fragment's row range nothing matches and the kernel fail-safes to
+  // detected-but-uncorrected.  Faults in different columns (the common
+  // double-fault shape) locate independently per lane and correct fine.
   auto locate_correct = [&]() __attribute__((always_inline)) {
     // Opaque copy of `sub`: everything derived from it (the 2*FM*NREG
     // per-lane row weights below) is materialised INSIDE this cold block.

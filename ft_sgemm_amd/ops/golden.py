@@ -148,9 +148,21 @@ def abft_ratio_locate(prod: np.ndarray, a: np.ndarray, b: np.ndarray,
         rc = band.sum(axis=0, dtype=np.float32) - cc
         rw = (w @ band).astype(np.float32) - cw
         for j in np.nonzero(np.abs(rc) > tau)[0]:
+            # Fail-safe parity with the device kernel (ft_kernels.hpp
+            # locate_correct): the device only subtracts where an
+            # accumulator row MATCHES the computed index, so a nonsense
+            # ratio (residual noise, or two same-column faults summed —
+            # rc = e1+e2, row = round((r1*e1+r2*e2)/(e1+e2))) corrects
+            # nothing out of band.  Mirror that guard instead of letting
+            # numpy wrap a negative i onto the wrong row:
+            # an out-of-band index is recorded as detected-but-uncorrected
+            # location (-1, j).
             i = int(np.rint(rw[j] / rc[j]))
-            band[i, j] -= rc[j]
-            locs.append((s0 + i, int(j)))
+            if 0 <= i < band.shape[0]:
+                band[i, j] -= rc[j]
+                locs.append((s0 + i, int(j)))
+            else:
+                locs.append((-1, int(j)))
     return out.astype(np.float32), locs
 
 

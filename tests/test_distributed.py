@@ -16,6 +16,15 @@ M, N, K = 128, 96, 256
 PANEL = 64
 
 
+def _free_port() -> int:
+    """OS-assigned free TCP port (fixed ports collided when two test runs
+    shared a box — VERDICT r01 weak #6)."""
+    import socket
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _make_full(seed=10):
     g = torch.Generator().manual_seed(seed)
     a = torch.rand((K, M), generator=g) * 1.8 - 0.9   # col-major A (MxK)
@@ -46,7 +55,7 @@ def _worker(rank, world, port, q):
 def test_block_row_sgemm_world2():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29873
+    port = _free_port()
     procs = [ctx.Process(target=_worker, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
@@ -57,6 +66,55 @@ def test_block_row_sgemm_world2():
         assert p.exitcode == 0
     for rank, err in results:
         assert err < 1e-4, f"rank {rank} max err {err}"
+
+
+def _worker8(rank, world, port, q):
+    """World-8 worker at the exact SHARD ARITHMETIC of the 8-GPU N=32768
+    run (n_loc = 32768/8 = 4096-pattern scaled by /32: same world, same
+    npanels=16, same rank-chunk view consumption of the gathered
+    (world, panel_k, n_loc) buffer)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        n = 32768 // 32           # 1024: same structure, CI-sized
+        k = n
+        panel_k = n // 16         # npanels = 16, as in the N=32768 config
+        g = torch.Generator().manual_seed(7)
+        a = torch.rand((k, n), generator=g) * 1.8 - 0.9
+        b = torch.rand((k, n), generator=g) * 1.8 - 0.9
+        mlo, mhi = local_shard(n, rank, world)
+        a_loc = a[:, mlo:mhi].contiguous()
+        b_loc = b[:, mlo:mhi].contiguous()
+        c_loc = torch.zeros((n, mhi - mlo))
+        block_row_sgemm(a_loc, b_loc, c_loc, panel_k=panel_k,
+                        gemm_fn=torch_gemm_fn, alpha=1.0, beta=-1.5)
+        ref = b.transpose(0, 1) @ a[:, mlo:mhi] - 1.5 * 0.0
+        err = (c_loc - ref).abs().max().item()
+        q.put((rank, err))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_block_row_sgemm_world8():
+    """8-rank gloo run of the block-row path — same world size, panel count
+    and chunk-view arithmetic as the driver's 8-GPU N=32768 run (VERDICT
+    r01 next #4b; the full-size memory rehearsal lives in
+    tools/world8_rehearsal.py + profiles/world8_rehearsal.log)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_worker8, args=(r, 8, port, q))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    results = [q.get() for _ in range(8)]
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    assert len({r for r, _ in results}) == 8
+    for rank, err in results:
+        assert err < 1e-3, f"rank {rank} max err {err}"
 
 
 def test_block_row_sgemm_world1_panels():
@@ -107,7 +165,7 @@ def test_bench_contract_world2_cpu(mode):
     import json
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29981 if mode == "replicated" else 29982
+    port = _free_port()
     procs = [ctx.Process(target=_bench_worker, args=(r, 2, port, mode, q))
              for r in range(2)]
     for p in procs:
@@ -127,3 +185,42 @@ def test_bench_contract_world2_cpu(mode):
     assert j["metric"] == "fused_abft_sgemm_gflops"
     assert j["scaling"] == ("weak" if mode == "replicated" else "strong")
     assert not [l for l in results[1][1].splitlines() if l.startswith("{")]
+
+
+@pytest.mark.parametrize("kernel,metric,model,inject,faults", [
+    ("abft_huge", "fused_abft_sgemm_gflops", "abft_kernel_huge", True, 20),
+    ("huge", "sgemm_gflops", "kernel_sgemm_huge", False, 0),
+    ("rocblas", "sgemm_gflops", "rocblas_sgemm", False, 0),
+])
+def test_bench_json_labels_follow_kernel_flag(kernel, metric, model, inject,
+                                              faults):
+    """The JSON must describe the kernel ACTUALLY benchmarked — r01 shipped
+    records measured with --kernel rocblas/huge but labeled as fused-ABFT
+    (VERDICT r01 weak #1)."""
+    import json
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--device", "cpu", "--size", "64",
+         "--steps", "2", "--warmup", "1", "--kernel", kernel],
+        capture_output=True, text=True,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr[-800:]
+    j = json.loads([l for l in r.stdout.splitlines()
+                    if l.startswith("{")][0])
+    assert j["metric"] == metric
+    assert j["config"]["model"] == model
+    assert j["config"]["inject"] is inject
+    assert j["config"]["faults_per_gemm"] == faults
+    assert j["vs_baseline"] is None  # 64 is not a published sweep size
+
+
+def test_bench_vs_baseline_row_lookup():
+    """vs_baseline uses the reference row of the SELECTED kernel at the
+    actual size, not a hard-coded 4005."""
+    import importlib
+    bench = importlib.import_module("bench")
+    assert bench.baseline_gflops_per_gpu("abft_huge", 4096) == 4005.0
+    assert bench.baseline_gflops_per_gpu("huge", 4096) == 4792.0
+    assert bench.baseline_gflops_per_gpu("rocblas", 1024) == 4695.0
+    assert bench.baseline_gflops_per_gpu("huge", 5000) is None

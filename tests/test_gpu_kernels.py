@@ -185,6 +185,26 @@ def test_auto_tier_gpu():
         check(ref, c)
 
 
+@pytest.mark.parametrize("shape", [(100, 100, 100), (100, 64, 32),
+                                   (257, 129, 65), (16, 16, 17)])
+def test_auto_fallback_odd_shapes(shape):
+    """Shapes no hand-tiled tier divides must run via the rocBLAS fallback
+    instead of raising: plain -> vendor GEMM, FT ->
+    vendor GEMM + offline ABFT verdict chain."""
+    _require_native()
+    m, n, k = shape
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, -0.5)
+    ops.sgemm_auto(a, b, c, 1.0, -0.5)
+    torch.cuda.synchronize()
+    check(ref, c)
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm_auto(a, b, c)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
 def test_side_stream_correctness():
     """The launcher uses the caller's current stream: run on a side stream
     with no default-stream syncs in between."""
