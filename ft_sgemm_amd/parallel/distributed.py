@@ -79,7 +79,10 @@ def block_row_sgemm(a_local: torch.Tensor, b_local: torch.Tensor,
     n_loc = b_local.shape[1]
     npanels = k // panel_k
 
-    if world == 1:
+    # An INITIALIZED world-1 group still takes the collective path below —
+    # that is how the RCCL branch (device all_gather_into_tensor consumed by
+    # MFMA kernels) is exercised on a single-GPU box (VERDICT r01 next #4a).
+    if world == 1 and not dist.is_initialized():
         for p in range(npanels):
             sl = slice(p * panel_k, (p + 1) * panel_k)
             gemm_fn(a_local[sl].contiguous(), b_local[sl].contiguous(),
