@@ -21,7 +21,8 @@ bin/ft_sgemm: csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip csrc/ft_
 ext:
 	# ninja does not track header deps for hipcc sources: rebuild .hip TUs
 	# whenever a csrc header is newer than the built extension
-	@if [ -n "$$(find csrc -name '*.hpp' -o -name '*.h' -newer ft_sgemm_amd/_C.cpython-310-x86_64-linux-gnu.so 2>/dev/null)" ]; then touch csrc/*.hip csrc/generated/*.hip; fi
+	@SO=$$(ls -t ft_sgemm_amd/_C*.so 2>/dev/null | head -1); \
+	if [ -n "$$SO" ] && [ -n "$$(find csrc \( -name '*.hpp' -o -name '*.h' \) -newer $$SO 2>/dev/null)" ]; then touch csrc/*.hip csrc/generated/*.hip; fi
 	PYTORCH_ROCM_ARCH=$(ARCH) python3 setup.py build_ext --inplace
 
 clean:

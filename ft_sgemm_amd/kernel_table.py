@@ -33,13 +33,18 @@ TILING = OrderedDict(
     medium=dict(bm=32, bn=32, bk=16, wm=32, wn=32, mfma="f32_32x32x2"),
     # large: 2 waves of 32x64 beat the single 64x64 wave by ~6% plain and
     # ~11% fused at N=4096 (probe_pipeline PROBE_ONLY=T)
-    large=dict(bm=64, bn=64, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
+    large=dict(bm=64, bn=64, bk=16, wm=32, wn=64, mfma="f32_32x32x2",
+               streamk=True),
     tall=dict(bm=128, bn=32, bk=16, wm=64, wn=32, mfma="f32_32x32x2"),
     wide=dict(bm=32, bn=128, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
     # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32
     # at N=4096 (tools/probe_pipeline.hip T3/T6 vs PA) — bigger M-tile cuts
     # total A/B traffic 25% and the 48 KB LDS keeps 2 blocks/CU resident.
-    huge=dict(bm=256, bn=128, bk=16, wm=128, wn=64, mfma="f32_32x32x2"),
+    # streamk: also build the stream-K twin (csrc/ft_streamk.hpp) — the
+    # launcher auto-selects it at grid-straggler sizes where the classic
+    # tile-per-workgroup grid leaves a tail dispatch round mostly idle.
+    huge=dict(bm=256, bn=128, bk=16, wm=128, wn=64, mfma="f32_32x32x2",
+              streamk=True),
 )
 
 TIERS = list(TILING.keys())

@@ -1,0 +1,116 @@
+"""Stream-K kernel correctness (csrc/ft_streamk.hpp): the work-centric
+decomposition that fixes the grid-straggler sweep sizes (VERDICT r01 next
+#3).  FT_SGEMM_STREAMK=1 forces the stream-K path so these tests pin it
+regardless of the auto heuristic; every case is checked against the plain
+torch fp32 reference, the fused cases with the always-on 20-fault
+injector (split tiles must verify+correct their PARTIAL accumulations)."""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from ft_sgemm_amd import ops
+
+ABS_TOL = 1e-2
+REL_TOL = 1e-2
+
+
+@pytest.fixture(autouse=True)
+def force_streamk():
+    os.environ["FT_SGEMM_STREAMK"] = "1"
+    yield
+    os.environ.pop("FT_SGEMM_STREAMK", None)
+
+
+def check(ref, got):
+    diff = (ref - got).abs()
+    rel = diff / ref.abs().clamp_min(1e-30)
+    bad = (diff > ABS_TOL) & (rel > REL_TOL)
+    assert not bad.any(), (
+        f"{int(bad.sum())} mismatches, max abs diff {diff.max().item():.4e}")
+
+
+# shapes chosen to cover: tiles < G (full split regime), a straggler grid,
+# single-tile (all units in one tile), and K smaller than one 64-k unit
+# per split boundary behaviour
+SHAPES = [(1024, 1024, 1024), (3072, 3072, 3072), (256, 128, 2048),
+          (512, 384, 640), (2048, 2048, 192)]
+
+
+@pytest.mark.parametrize("tier", ["huge", "large"])
+@pytest.mark.parametrize("shape", SHAPES)
+def test_streamk_plain(tier, shape):
+    m, n, k = shape
+    if any(d % t != 0 for d, t in
+           [(m, 256 if tier == "huge" else 64),
+            (n, 128 if tier == "huge" else 64), (k, 64)]):
+        pytest.skip("shape does not divide this tier")
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.sgemm(tier, a, b, c, 1.0, 0.0)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+@pytest.mark.parametrize("tier", ["huge", "large"])
+@pytest.mark.parametrize("shape", SHAPES)
+def test_streamk_fused_inject(tier, shape):
+    m, n, k = shape
+    if any(d % t != 0 for d, t in
+           [(m, 256 if tier == "huge" else 64),
+            (n, 128 if tier == "huge" else 64), (k, 64)]):
+        pytest.skip("shape does not divide this tier")
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm(tier, a, b, c, 1.0, 0.0, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_streamk_beta_prescale():
+    """beta is applied by the one-pass prescale kernel exactly once, then
+    split tiles accumulate atomically."""
+    m = n = k = 1024
+    a, b, c = ops.make_operands(m, n, k)
+    c.normal_()
+    c0 = c.clone()
+    ref = ops.torch_reference(a, b, c0, 1.0, -1.5)
+    ops.ft_sgemm("huge", a, b, c, 1.0, -1.5, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_streamk_off_env():
+    """FT_SGEMM_STREAMK=0 must take the classic path (same numerics)."""
+    os.environ["FT_SGEMM_STREAMK"] = "0"
+    m = n = k = 1024
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_streamk_k_not_multiple_of_64_falls_back():
+    """K % 64 != 0 is outside the stream-K unit quantum: the wrapper must
+    fall back to the classic kernel even when forced."""
+    m, n, k = 512, 256, 80  # k % 16 == 0 but k % 64 != 0
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True)
+    torch.cuda.synchronize()
+    check(ref, c)
+
+
+def test_streamk_high_fault_rate():
+    """Dense verify windows on split tiles: istride drops to 1 (one
+    inject+verify per 64-k window per tile segment)."""
+    m = n = k = 2048
+    a, b, c = ops.make_operands(m, n, k)
+    ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+    ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True, verify_windows=64)
+    torch.cuda.synchronize()
+    check(ref, c)
