@@ -10,7 +10,8 @@ gen:
 
 cli: bin/ft_sgemm
 
-KERNEL_TUS := $(wildcard csrc/generated/kernel_*.hip)
+# filter out PyTorch-hipify byproducts (kernel_*_hip.hip copies)
+KERNEL_TUS := $(filter-out %_hip.hip, $(wildcard csrc/generated/kernel_*.hip))
 
 bin/ft_sgemm: csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip csrc/ft_kernels.hpp csrc/tier_launch.hpp csrc/ft_core.h csrc/generated/tile_params.h $(KERNEL_TUS)
 	mkdir -p bin

@@ -139,7 +139,9 @@ int main(int argc, char** argv) {
   HIP_CALL(hipMalloc(&ws_sb, maxn * sizeof(float)));
   HIP_CALL(hipMalloc(&ws_rrow, maxn * sizeof(float)));
   HIP_CALL(hipMalloc(&ws_rcol, maxn * sizeof(float)));
-  HIP_CALL(hipMalloc(&ws_dres, 2 * sizeof(float)));
+  // verdict slot pairs: one per verified panel (worst-panel semantics);
+  // panel_k >= 64 always, so maxn/64 bounds the panel count
+  HIP_CALL(hipMalloc(&ws_dres, 2 * (maxn / 64 + 2) * sizeof(float)));
   {
     std::vector<float> ones(maxn, 1.f);
     HIP_CALL(hipMemcpy(ws_ones, ones.data(), maxn * sizeof(float),

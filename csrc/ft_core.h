@@ -34,7 +34,9 @@ struct BaselineWorkspace {
   float* s_b;      // panel_k (B-panel row sums)
   float* ref_row;  // M   (maintained row checksum)
   float* ref_col;  // N   (maintained col checksum)
-  float* d_res;    // 2   (device-side dot verdicts, stream-ordered)
+  float* d_res;    // 2 * ceil(K / panel_k)  (device-side dot verdict slot
+                   // pair per verified panel, stream-ordered; the host
+                   // reduces to the WORST panel's verdicts)
 };
 
 int rocblas_sgemm_nt(int M, int N, int K, const float* A, const float* B,

@@ -7,6 +7,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdio>
 #include <cstdlib>
 
 #include "ft_kernels.hpp"
@@ -63,10 +64,15 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
     kfn = (const void*)&sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, false,
                                            false>;
   int maxblk = 0;
-  if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&maxblk, kfn, THREADS,
-                                                   0) != hipSuccess ||
-      maxblk < 1)
+  const hipError_t oe =
+      hipOccupancyMaxActiveBlocksPerMultiprocessor(&maxblk, kfn, THREADS, 0);
+  const bool dbg = std::getenv("FT_SGEMM_SK_DEBUG") != nullptr;
+  if (oe != hipSuccess || maxblk < 1) {
+    if (dbg)
+      fprintf(stderr, "[sk %dx%d] occupancy query failed: err=%d maxblk=%d\n",
+              BM, BN, (int)oe, maxblk);
     return hipErrorNotSupported;
+  }
 
   const int tiles = (M / BM) * (N / BN);
   int G0 = device_cu_count() * maxblk;
@@ -76,10 +82,20 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (mode != 1) {
     const float rounds = ceilf((float)tiles / (float)G0);
     const float waste = (rounds * G0 - tiles) / (rounds * G0);
-    if (waste < 0.15f) return hipErrorNotSupported;
+    if (waste < 0.15f) {
+      if (dbg)
+        fprintf(stderr, "[sk %dx%d] M=%d N=%d: classic (waste %.3f)\n", BM,
+                BN, M, N, waste);
+      return hipErrorNotSupported;
+    }
   }
   const int total = tiles * (K >> 6);
   const int G = total < G0 ? total : G0;
+  if (dbg)
+    fprintf(stderr,
+            "[sk %dx%d] M=%d N=%d K=%d: ENGAGED tiles=%d maxblk=%d G=%d "
+            "units/wg~%.1f\n",
+            BM, BN, M, N, K, tiles, maxblk, G, (float)total / G);
 
   // beta applied once up front (split tiles accumulate into C)
   const size_t total4 = (size_t)M * N / 4;

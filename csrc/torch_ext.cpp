@@ -80,7 +80,10 @@ std::tuple<double, double> baseline_ft(at::Tensor a, at::Tensor b,
   at::Tensor row_c = at::empty({M}, opts), col_c = at::empty({N}, opts);
   at::Tensor s_a = at::empty({panel_k}, opts), s_b = at::empty({panel_k}, opts);
   at::Tensor ref_row = at::empty({M}, opts), ref_col = at::empty({N}, opts);
-  at::Tensor d_res = at::empty({2}, opts);
+  // one verdict slot pair per (potentially) verified panel (worst-panel
+  // semantics)
+  int64_t npanels = (K + panel_k - 1) / panel_k;
+  at::Tensor d_res = at::empty({2 * npanels}, opts);
   ftsgemm::BaselineWorkspace ws{
       ones.mutable_data_ptr<float>(),    row_c.mutable_data_ptr<float>(),
       col_c.mutable_data_ptr<float>(),   s_a.mutable_data_ptr<float>(),

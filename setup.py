@@ -28,7 +28,8 @@ setup(
                 "csrc/torch_ext.cpp",
                 "csrc/dispatch.hip",
                 "csrc/rocblas_path.hip",
-            ] + sorted(glob.glob("csrc/generated/kernel_*.hip")),
+            ] + sorted(p for p in glob.glob("csrc/generated/kernel_*.hip")
+                       if not p.endswith("_hip.hip")),
             include_dirs=[os.path.join(ROOT, "csrc")],
             libraries=["rocblas"],
             extra_compile_args={

@@ -113,6 +113,35 @@ def test_baseline_ft_verdict_and_result():
     assert res_row < 1.0 and res_col < 1.0
 
 
+@pytest.mark.parametrize("env", [
+    {},                                             # fast custom kernels
+    {"FT_SGEMM_BASELINE_MODE": "chain"},            # strict rocBLAS chain
+    {"FT_SGEMM_VERIFY_EVERY": "2"},                 # verdict cadence knob
+])
+def test_baseline_ft_modes(env):
+    """id-10 modes agree: custom reduction kernels vs the pure rocBLAS
+    chain, and the j-panel verdict cadence (always verifies the last
+    panel)."""
+    import os
+    _require_native()
+    m, n, k = 512, 384, 3072  # odd panel tail: 3072 = 2x1024 + 1024
+    for kk, vv in env.items():
+        os.environ[kk] = vv
+    try:
+        a, b, c = ops.make_operands(m, n, k)
+        ref = ops.torch_reference(a, b, c, 1.0, -1.5)
+        c.normal_()
+        ref = ops.torch_reference(a, b, c, 1.0, -1.5)
+        _, (res_row, res_col) = ops.baseline_ft(a, b, c, 1.0, -1.5,
+                                                panel_k=1024)
+        torch.cuda.synchronize()
+        check(ref, c)
+        assert res_row < 1.0 and res_col < 1.0
+    finally:
+        for kk in env:
+            os.environ.pop(kk, None)
+
+
 def test_native_extension_is_loaded():
     """Guards against silent eager fallback: the .so must be in-tree."""
     import ft_sgemm_amd._C as ext

@@ -26,9 +26,15 @@ def log(msg):
 
 
 def time_gflops(fn, n, reps=5):
+    # steady-state warm: run the exact config for >=0.25 s so the clock ramp
+    # and allocator are out of the timed region (the r01 probe's single
+    # warm launch let run order skew A/B by up to 10%)
     torch.cuda.synchronize()
-    fn()  # warm
-    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < 0.25:
+        fn()
+        torch.cuda.synchronize()
+    reps = max(reps, int(200e9 / (2.0 * n * n * n)))  # >=~2ms timed
     beg = torch.cuda.Event(enable_timing=True)
     end = torch.cuda.Event(enable_timing=True)
     beg.record()
@@ -87,6 +93,23 @@ def main():
         del a, b, c
         torch.cuda.empty_cache()
     os.environ.pop("FT_SGEMM_STREAMK", None)
+
+    # one debug launch per (size, tier) so the log records whether the
+    # auto heuristic engaged and with what G (stderr from the extension)
+    os.environ["FT_SGEMM_SK_DEBUG"] = "1"
+    for n in sizes:
+        for tier in tiers:
+            bm = 256 if tier == "huge" else 64
+            bn = 128 if tier == "huge" else 64
+            if n % bm or n % bn or n % 64:
+                continue
+            a, b, c = ops.make_operands(n, n, n)
+            ops.sgemm(tier, a, b, c, 1.0, 0.0)
+            torch.cuda.synchronize()
+            del a, b, c
+            torch.cuda.empty_cache()
+    os.environ.pop("FT_SGEMM_SK_DEBUG", None)
+
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/sk_probe.log", "w") as f:
         f.write("\n".join(LINES) + "\n")
