@@ -43,6 +43,7 @@
 
 namespace ftsgemm {
 
+using f32x2 = __attribute__((ext_vector_type(2))) float;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 
@@ -97,17 +98,22 @@ __device__ constexpr int acc_row(int reg, int sub) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * sub;
 }
 
-// Segment-sum precompute: SA[seg * sstr + k] = sum_{i in [seg*SEG, +SEG)}
-// A[i + k*M] for a column-major MxK matrix, and (when SW != nullptr) the
-// row-index-weighted sums SW[seg * sstr + k] = sum_i (i % SEG) * A[i + k*M]
-// used by the ratio locate (fault row = round(rw / rc)).  One workgroup per
-// column k, coalesced f32x4 sweep, SEG/4-lane shuffle groups, no LDS, no
-// atomics.  sstr must be >= K and a multiple of 64 (the fused kernel
-// streams 64-k strips with a 4-B/lane global_load_lds, never crossing rows).
+// Segment-sum precompute: for a column-major MxK matrix, per SEG-row band,
+// the plain column sum s = sum_{i in [seg*SEG, +SEG)} A[i + k*M] and the
+// row-index-weighted sum w = sum_i (i % SEG) * A[i + k*M] used by the
+// ratio locate (fault row = round(rw / rc)).  The pair is written
+// INTERLEAVED — S[seg * 2*sstr + 2k] = s, S[seg * 2*sstr + 2k + 1] = w —
+// so the fused kernel's encode fetches both with a single ds_read_b64 per
+// k-step (the two separate b32 broadcast reads were one of the extra
+// issue slots behind the tall tier's 21% fused overhead).
+// One workgroup per column k, coalesced f32x4 sweep, SEG/4-lane shuffle
+// groups, no LDS, no atomics.  sstr must be >= K and a multiple of 64 (the
+// fused kernel streams 64-k strips with 4-B/lane global_load_lds pairs,
+// never crossing rows).
 template <int SEG>
 __global__ __launch_bounds__(256) void segsum_kernel(
     int M, int K, int sstr, const float* __restrict__ A,
-    float* __restrict__ SA, float* __restrict__ SW) {
+    float* __restrict__ S) {
   const int k = blockIdx.x;
   const int tid = threadIdx.x;
   const float* col = A + (size_t)k * M;
@@ -128,8 +134,8 @@ __global__ __launch_bounds__(256) void segsum_kernel(
       w += __shfl_xor(w, m, 64);
     }
     if (idx < M && (idx % SEG) == 0) {
-      SA[(size_t)(idx / SEG) * sstr + k] = s;
-      if (SW) SW[(size_t)(idx / SEG) * sstr + k] = w;
+      S[(size_t)(idx / SEG) * 2 * sstr + 2 * k] = s;
+      S[(size_t)(idx / SEG) * 2 * sstr + 2 * k + 1] = w;
     }
   }
 }
@@ -235,26 +241,25 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
 
   // ---- ABFT strip staging: one 4-B/lane glds pair per TWO K panels ----
   // Each wave streams its own 64-k window of the precomputed segment sums
-  // (SA row of its WM range, SB row of its WN range) into a private LDS
-  // strip.  No extra barrier, no cooperative pass: the strips ride the
-  // same prefetch pipeline as the A/B panels and are drained by the same
-  // end-of-panel __syncthreads.
-  // Workspace layout (set up by the launcher): SA = [plain A-segment sums
-  // (M/WM rows) | row-weighted A-segment sums (M/WM rows)]; each row sstr
-  // floats.  SB is unused by this scheme and passed null.
+  // into a private LDS strip.  No extra barrier, no cooperative pass: the
+  // strips ride the same prefetch pipeline as the A/B panels and are
+  // drained by the same end-of-panel __syncthreads.
+  // Workspace layout (set up by the launcher): per WM-row band of A, one
+  // row of 2*sstr floats with (plain, row-weighted) segment sums
+  // INTERLEAVED at [2k], [2k+1] — a 64-k window is 128 contiguous floats,
+  // fetched by two 64-lane glds loads and consumed with one ds_read_b64
+  // per k-step.
   const int segA = bx * WAVES_M + wm_idx;
-  const int segsA = ABFT ? (M / WM) : 0;
   auto strip_stage = [&](int pb, int k0) __attribute__((always_inline)) {
     // glds source addresses are PER-LANE (the LDS side is uniform base +
-    // lane*size): lane l fetches SA[segA][k0 + l] into strip slot l.
-    const float* ga = SA + (size_t)segA * sstr + k0 + lane;
-    const float* gw = SA + (size_t)(segsA + segA) * sstr + k0 + lane;
+    // lane*size): lane l fetches pair-element (2*k0 + l).
+    const float* ga = SA + (size_t)segA * 2 * sstr + 2 * k0 + lane;
     float* dst = &lds[STRIP_OFF + wave * 256 + pb * 128];
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)ga,
         (__attribute__((address_space(3))) void*)dst, 4, 0, 0);
     __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)gw,
+        (const __attribute__((address_space(1))) void*)(ga + 64),
         (__attribute__((address_space(3))) void*)(dst + 64), 4, 0, 0);
   };
 
@@ -424,7 +429,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
       const float* Bs = &lds[q * BUF + BM * BK];
       const float* strip =
           ABFT ? &lds[STRIP_OFF + wave * 256 + ((it / PPS) & 1) * 128 +
-                      (it % PPS) * BK]
+                      (it % PPS) * BK * 2]
                : nullptr;
 #pragma unroll
       for (int kk = 0; kk < BK / KSTEP; ++kk) {
@@ -438,17 +443,16 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
           b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
 
         if constexpr (ABFT) {
-          // Encode: two broadcast LDS reads of the precomputed segment
-          // sums (plain + row-weighted) for this k-slice + two fmas per
-          // B fragment into the running column checksums (reference
+          // Encode: ONE ds_read_b64 fetches the interleaved (plain,
+          // row-weighted) segment-sum pair for this k-slice + two fmas
+          // per B fragment into the running column checksums (reference
           // encode: ft_sgemm_huge.cuh:150-213, redesigned around the
           // offline segsum pass + ratio locate).
-          const float sa = strip[kloc];
-          const float saw = strip[64 + kloc];
+          const f32x2 sw = *(const f32x2*)(strip + 2 * kloc);
 #pragma unroll
           for (int fn = 0; fn < FN; ++fn) {
-            cc[fn] = fmaf(sa, b[fn], cc[fn]);
-            cw[fn] = fmaf(saw, b[fn], cw[fn]);
+            cc[fn] = fmaf(sw[0], b[fn], cc[fn]);
+            cw[fn] = fmaf(sw[1], b[fn], cw[fn]);
           }
         }
 

@@ -9,18 +9,31 @@
 // every sweep size over-fills it), but its ratio-to-vendor-BLAS is >=100%
 // at EVERY size (README.md:38-53), so ours must not crater.
 //
-// Design (work-centric, CUTLASS-stream-K-like but simplified around the
-// ABFT strip windows): the GEMM is a list of `total = ntiles * (K/64)`
-// work units — one unit = one 64-k strip window of one output tile (4
-// BK=16 LDS panels, exactly the granularity the ABFT checksum strips are
-// staged at).  A fixed grid of G workgroups (G = CUs x occupancy, never
-// more than total) each processes a CONTIGUOUS unit range, k-fastest:
-// perfect load balance at any size, no tail round.  A tile fully owned by
-// one workgroup takes the normal alpha/beta epilogue; a tile split across
-// workgroups is combined with native f32 global atomics
-// (global_atomic_add_f32 via unsafeAtomicAdd) after a one-pass C prescale
-// kernel has applied beta — fp32 add-order nondeterminism across splits is
-// far below the 1e-2/1e-2 verify tolerance at the reference operand scale.
+// Design (work-centric, CUTLASS-stream-K-like but built around the ABFT
+// strip windows): the GEMM is a list of `total = ntiles * (K/64)` work
+// units — one unit = one 64-k strip window of one output tile (4 BK=16
+// LDS panels, exactly the granularity the ABFT checksum strips are staged
+// at).  A fixed grid of G workgroups (G = CUs x occupancy, never more
+// than total — ALL workgroups are co-resident, which the fixup protocol
+// relies on) each processes a CONTIGUOUS unit range, k-fastest: perfect
+// load balance at any size, no tail round.
+//
+// Split-tile combine is the deterministic owner-fixup protocol, not
+// atomics (an atomic f32 combine was measured 3x SLOWER than the classic
+// straggler grid at N=1024 — 16 workgroups hammering the same 128 KB of
+// C serialize in L2):
+//   * a tile fully inside one workgroup's range takes the normal
+//     alpha/beta epilogue directly;
+//   * a workgroup whose range STARTS mid-tile (only its first segment
+//     can) writes its raw partial accumulator to a private workspace
+//     slot — coalesced streaming stores, no contention — and
+//     release-stores a per-workgroup flag;
+//   * the tile's OWNER (the workgroup holding the tile's first unit)
+//     acquire-spins on the flags of the (consecutive) contributor
+//     workgroups, adds their partials from the workspace, and runs the
+//     normal alpha/beta epilogue.  Owners only ever wait on HIGHER
+//     workgroup ids, so the dependency graph is acyclic, and summation
+//     order is deterministic.
 //
 // ABFT composes cleanly: each workgroup's per-lane column checksums (cc,
 // cw) cover exactly the k-range it accumulated, so the wave-autonomous
@@ -35,35 +48,15 @@
 
 namespace ftsgemm {
 
-// C *= beta in one grid-stride f32x4 pass (beta==0 zero-fills: stream-K
-// split tiles ACCUMULATE into C, so beta must be applied up front exactly
-// once — the classic kernels instead fold beta into their epilogue).
-// static: per-TU copy (only the stream-K TUs instantiate it); a plain
-// external __global__ here would collide at link across kernel_*_sk.hip.
-static __global__ __launch_bounds__(256) void prescale_kernel(
-    size_t total4, float beta, float* __restrict__ C) {
-  const size_t stride = (size_t)gridDim.x * 256;
-  for (size_t i = (size_t)blockIdx.x * 256 + threadIdx.x; i < total4;
-       i += stride) {
-    f32x4* p = (f32x4*)C + i;
-    if (beta == 0.f) {
-      *p = f32x4{0.f, 0.f, 0.f, 0.f};
-    } else {
-      f32x4 v = *p;
-#pragma unroll
-      for (int u = 0; u < 4; ++u) v[u] *= beta;
-      *p = v;
-    }
-  }
-}
-
 template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
           bool INJECT, int OCC = 2>
 __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC)
 void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
                         const float* __restrict__ B, float* __restrict__ C,
-                        float alpha, int istride, float tau, float inj_mag,
-                        const float* __restrict__ SA, int sstr) {
+                        float alpha, float beta, int istride, float tau,
+                        float inj_mag, const float* __restrict__ SA, int sstr,
+                        float* __restrict__ partials,
+                        unsigned* __restrict__ flags) {
   using T = mfma_traits<MM>;
   constexpr int KSTEP = T::kstep;
   constexpr int NREG = T::nreg;
@@ -131,17 +124,16 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
     }
   };
 
-  const int segsA = ABFT ? (M / WM) : 0;
+  // interleaved (plain, weighted) segment-sum pairs: see ft_kernels.hpp
   auto strip_stage = [&](int pb, int k0, int segA)
       __attribute__((always_inline)) {
-    const float* ga = SA + (size_t)segA * sstr + k0 + lane;
-    const float* gw = SA + (size_t)(segsA + segA) * sstr + k0 + lane;
+    const float* ga = SA + (size_t)segA * 2 * sstr + 2 * k0 + lane;
     float* dst = &lds[STRIP_OFF + wave * 256 + pb * 128];
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)ga,
         (__attribute__((address_space(3))) void*)dst, 4, 0, 0);
     __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)gw,
+        (const __attribute__((address_space(1))) void*)(ga + 64),
         (__attribute__((address_space(3))) void*)(dst + 64), 4, 0, 0);
   };
 
@@ -237,7 +229,7 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
         const float* Bs = &lds[qb * BUF + BM * BK];
         const float* strip =
             ABFT ? &lds[STRIP_OFF + wave * 256 + ((w_lo + p / PPS) & 1) * 128 +
-                        (p % PPS) * BK]
+                        (p % PPS) * BK * 2]
                  : nullptr;
 #pragma unroll
         for (int kk = 0; kk < BK / KSTEP; ++kk) {
@@ -250,12 +242,11 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
           for (int fn = 0; fn < FN; ++fn)
             b[fn] = Bs[kloc * BN + wj0 + fn * MM + r];
           if constexpr (ABFT) {
-            const float sa = strip[kloc];
-            const float saw = strip[64 + kloc];
+            const f32x2 sw = *(const f32x2*)(strip + 2 * kloc);
 #pragma unroll
             for (int fn = 0; fn < FN; ++fn) {
-              cc[fn] = fmaf(sa, b[fn], cc[fn]);
-              cw[fn] = fmaf(saw, b[fn], cw[fn]);
+              cc[fn] = fmaf(sw[0], b[fn], cc[fn]);
+              cw[fn] = fmaf(sw[1], b[fn], cw[fn]);
             }
           }
           __builtin_amdgcn_iglp_opt(0);
@@ -270,33 +261,80 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
       if constexpr (ABFT) verify_correct();
     }
 
-    // ---- tile contribution ----
-    // full owner: C was pre-scaled by beta, sole writer -> plain RMW;
-    // split tile: accumulate with native f32 global atomics.
-    const bool full = (w_lo == 0) && (w_hi == upt);
+    // ---- tile contribution (owner-fixup protocol, see header) ----
+    constexpr int TPT = FM * FN * NREG;  // per-thread floats of one tile
+    if (w_lo != 0) {
+      // contributor: only the FIRST segment of a workgroup can start
+      // mid-tile.  Raw partial -> private slot, then release the flag.
+      float* slot = partials + (size_t)g * (BM * BN) + tid * TPT;
 #pragma unroll
-    for (int fm = 0; fm < FM; ++fm)
+      for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < FN; ++fn) {
-        const int j = jn0 + wj0 + fn * MM + r;
-        float* colbase = C + (size_t)j * M + im0 + wi0 + fm * MM;
+        for (int fn = 0; fn < FN; ++fn)
 #pragma unroll
-        for (int g4 = 0; g4 < NREG / 4; ++g4) {
-          float* p4 = colbase + 4 * sub + 8 * g4;
-          if (full) {
-            const f32x4 prev = *(const f32x4*)p4;
-            f32x4 out;
+          for (int g4 = 0; g4 < NREG / 4; ++g4) {
+            f32x4 v;
 #pragma unroll
-            for (int v = 0; v < 4; ++v)
-              out[v] = fmaf(alpha, acc[fm][fn][4 * g4 + v], prev[v]);
-            *(f32x4*)p4 = out;
-          } else {
-#pragma unroll
-            for (int v = 0; v < 4; ++v)
-              unsafeAtomicAdd(p4 + v, alpha * acc[fm][fn][4 * g4 + v]);
+            for (int q4 = 0; q4 < 4; ++q4) v[q4] = acc[fm][fn][4 * g4 + q4];
+            *(f32x4*)(slot + (fm * FN + fn) * NREG + 4 * g4) = v;
           }
+      __threadfence();   // each thread's stores visible agent-wide ...
+      __syncthreads();   // ... before thread 0 publishes for the group
+      if (tid == 0)
+        __hip_atomic_store(&flags[g], 1u, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    } else {
+      if (w_hi != upt) {
+        // owner of a split tile: fold in every contributor (consecutive
+        // higher workgroup ids whose ranges start inside this tile)
+        const int last_u = tile * upt + upt - 1;
+        int gl;  // workgroup owning the tile's last unit
+        if (last_u < rem * (q + 1)) gl = last_u / (q + 1);
+        else gl = rem + (last_u - rem * (q + 1)) / q;
+        for (int gc = g + 1; gc <= gl; ++gc) {
+          while (__hip_atomic_load(&flags[gc], __ATOMIC_ACQUIRE,
+                                   __HIP_MEMORY_SCOPE_AGENT) == 0u)
+            __builtin_amdgcn_s_sleep(8);
+          const float* slot = partials + (size_t)gc * (BM * BN) + tid * TPT;
+#pragma unroll
+          for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+            for (int fn = 0; fn < FN; ++fn)
+#pragma unroll
+              for (int g4 = 0; g4 < NREG / 4; ++g4) {
+                const f32x4 v =
+                    *(const f32x4*)(slot + (fm * FN + fn) * NREG + 4 * g4);
+#pragma unroll
+                for (int q4 = 0; q4 < 4; ++q4)
+                  acc[fm][fn][4 * g4 + q4] += v[q4];
+              }
         }
       }
+      // normal alpha/beta epilogue (identical to the classic kernel)
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn) {
+          const int j = jn0 + wj0 + fn * MM + r;
+          float* colbase = C + (size_t)j * M + im0 + wi0 + fm * MM;
+#pragma unroll
+          for (int g4 = 0; g4 < NREG / 4; ++g4) {
+            float* p4 = colbase + 4 * sub + 8 * g4;
+            f32x4 out;
+            if (beta != 0.f) {
+              const f32x4 prev = *(const f32x4*)p4;
+#pragma unroll
+              for (int q4 = 0; q4 < 4; ++q4)
+                out[q4] = alpha * acc[fm][fn][4 * g4 + q4] + beta * prev[q4];
+            } else {
+#pragma unroll
+              for (int q4 = 0; q4 < 4; ++q4)
+                out[q4] = alpha * acc[fm][fn][4 * g4 + q4];
+            }
+            *(f32x4*)p4 = out;
+          }
+        }
+    }
     u += seg_units;
   }
 }

@@ -97,12 +97,12 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
             "units/wg~%.1f\n",
             BM, BN, M, N, K, tiles, maxblk, G, (float)total / G);
 
-  // beta applied once up front (split tiles accumulate into C)
-  const size_t total4 = (size_t)M * N / 4;
-  const int pgrid =
-      (int)((total4 + 255) / 256 < 8192 ? (total4 + 255) / 256 : 8192);
-  hipLaunchKernelGGL(prescale_kernel, dim3(pgrid), dim3(256), 0, stream,
-                     total4, beta, C);
+  // The owner-fixup workspace is stream-ordered (concurrent GEMMs on
+  // different streams must not share it); hipMallocAsync is not legal
+  // inside graph capture, so captured launches take the classic path.
+  hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+  (void)hipStreamIsCapturing(stream, &cap);
+  if (cap != hipStreamCaptureStatusNone) return hipErrorNotSupported;
 
   const float* SA = nullptr;
   int sstr = 0;
@@ -110,7 +110,7 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
     if (!ws) return hipErrorInvalidValue;
     sstr = abft_sstr(K);
     hipLaunchKernelGGL((segsum_kernel<WM>), dim3(K), dim3(256), 0, stream, M,
-                       K, sstr, A, ws, ws + (size_t)(M / WM) * sstr);
+                       K, sstr, A, ws);
     SA = ws;
   }
   // inject+verify cadence: one pass per `istride` strip windows, targeting
@@ -119,28 +119,44 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   int istride = upt / (verify_windows > 0 ? verify_windows : 20);
   if (istride < 1) istride = 1;
 
-  if (abft && inject) {
-    hipLaunchKernelGGL(
-        (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true, true>), dim3(G),
-        dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, istride, tau,
-        inj_mag, SA, sstr);
-  } else if (abft) {
-    hipLaunchKernelGGL(
-        (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true, false>), dim3(G),
-        dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, istride, tau,
-        inj_mag, SA, sstr);
-  } else {
-    hipLaunchKernelGGL(
-        (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, false, false>), dim3(G),
-        dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, istride, tau,
-        inj_mag, SA, sstr);
+  // partial-tile slots (one per workgroup) + per-workgroup flags
+  float* partials = nullptr;
+  unsigned* flags = nullptr;
+  const size_t pbytes = (size_t)G * BM * BN * sizeof(float);
+  if (hipMallocAsync((void**)&partials, pbytes + G * sizeof(unsigned),
+                     stream) != hipSuccess)
+    return hipErrorNotSupported;  // nothing mutated yet; classic path
+  flags = (unsigned*)(partials + (size_t)G * BM * BN);
+  hipError_t err = hipMemsetAsync(flags, 0, G * sizeof(unsigned), stream);
+
+  if (err == hipSuccess) {
+    if (abft && inject) {
+      hipLaunchKernelGGL(
+          (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true, true>), dim3(G),
+          dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
+          tau, inj_mag, SA, sstr, partials, flags);
+    } else if (abft) {
+      hipLaunchKernelGGL(
+          (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true, false>), dim3(G),
+          dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
+          tau, inj_mag, SA, sstr, partials, flags);
+    } else {
+      hipLaunchKernelGGL(
+          (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, false, false>), dim3(G),
+          dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
+          tau, inj_mag, SA, sstr, partials, flags);
+    }
+    err = hipGetLastError();
   }
-  return hipGetLastError();
+  (void)hipFreeAsync(partials, stream);
+  return err;
 }
 
-// ABFT workspace layout: [ (M/WM) rows of SA | (N/WN) rows of SB ], each row
-// sstr = round_up(K, 64) floats (the fused kernel streams 64-k strips with a
-// 4-B/lane global_load_lds and must not cross rows).
+// ABFT workspace layout: (M/WM) rows, one per WM-band of A, each of
+// 2 * sstr floats holding the (plain, row-weighted) segment-sum pairs
+// INTERLEAVED at [2k], [2k+1]; sstr = round_up(K, 64) so the fused
+// kernel's 64-k strip loads never cross rows.  B-side sums are not needed
+// by the ratio-locate scheme.
 inline int abft_sstr(int K) { return (K + 63) & ~63; }
 
 template <int WM, int WN>
@@ -175,7 +191,7 @@ hipError_t launch_tier(bool abft, bool inject, int M, int N, int K,
     if (!ws) return hipErrorInvalidValue;
     sstr = abft_sstr(K);
     hipLaunchKernelGGL((segsum_kernel<WM>), dim3(K), dim3(256), 0, stream, M,
-                       K, sstr, A, ws, ws + (size_t)(M / WM) * sstr);
+                       K, sstr, A, ws);
     SA = ws;
   }
   if (abft && inject) {
