@@ -137,21 +137,19 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
         (__attribute__((address_space(3))) void*)(dst + 64), 4, 0, 0);
   };
 
-  while (u < u_end) {
-    const int tile = u / upt;
-    const int w_lo = u - tile * upt;
-    const int seg_units = (u_end - u < upt - w_lo) ? (u_end - u)
-                                                   : (upt - w_lo);
-    const int w_hi = w_lo + seg_units;
-    const int bx = tile % ntm, by = tile / ntm;
-    const int im0 = bx * BM, jn0 = by * BN;
-    const int segA = bx * WAVES_M + wm_idx;
-    const int kbase = w_lo << 6;
-    const int npan = seg_units * PPS;
-
-    typename T::acc_t acc[FM][FN] = {};
-    float cc[FN] = {};
-    float cw[FN] = {};
+  // Accumulators and the ABFT verify machinery are declared ONCE, outside
+  // the tile loop, and re-zeroed per segment — declaring them inside the
+  // loop body made the allocator keep per-iteration copies alive and spill
+  // 142 VGPRs in the fused variants (measured; the plain variant fit).
+  typename T::acc_t acc[FM][FN];
+  float cc[FN];
+  float cw[FN];
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) acc[fm][fn] = {};
+#pragma unroll
+  for (int fn = 0; fn < FN; ++fn) cc[fn] = cw[fn] = 0.f;
 
     // verify/locate/correct: identical maths to the classic kernel
     // (csrc/ft_kernels.hpp locate_correct/verify_correct), operating on
@@ -200,6 +198,18 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
       for (int m = 1; m < 64; m <<= 1) res += __shfl_xor(res, m, 64);
       if (__builtin_expect(fabsf(res) > tau, 0)) locate_correct();
     };
+
+  while (u < u_end) {
+    const int tile = u / upt;
+    const int w_lo = u - tile * upt;
+    const int seg_units = (u_end - u < upt - w_lo) ? (u_end - u)
+                                                   : (upt - w_lo);
+    const int w_hi = w_lo + seg_units;
+    const int bx = tile % ntm, by = tile / ntm;
+    const int im0 = bx * BM, jn0 = by * BN;
+    const int segA = bx * WAVES_M + wm_idx;
+    const int kbase = w_lo << 6;
+    const int npan = seg_units * PPS;
 
     stage(0, kbase, im0, jn0);
     if constexpr (ABFT) strip_stage(w_lo & 1, kbase, segA);
@@ -292,9 +302,16 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
         if (last_u < rem * (q + 1)) gl = last_u / (q + 1);
         else gl = rem + (last_u - rem * (q + 1)) / q;
         for (int gc = g + 1; gc <= gl; ++gc) {
-          while (__hip_atomic_load(&flags[gc], __ATOMIC_ACQUIRE,
-                                   __HIP_MEMORY_SCOPE_AGENT) == 0u)
-            __builtin_amdgcn_s_sleep(8);
+          // one lane spins (256 threads x many owners hammering the same
+          // flag line was an L2 atomic storm); the barrier broadcasts the
+          // acquire to the workgroup (the acquiring lane's buffer_inv
+          // refreshes the CU-shared L1 all four waves read through)
+          if (tid == 0) {
+            while (__hip_atomic_load(&flags[gc], __ATOMIC_ACQUIRE,
+                                     __HIP_MEMORY_SCOPE_AGENT) == 0u)
+              __builtin_amdgcn_s_sleep(32);
+          }
+          __syncthreads();
           const float* slot = partials + (size_t)gc * (BM * BN) + tid * TPT;
 #pragma unroll
           for (int fm = 0; fm < FM; ++fm)
@@ -336,6 +353,16 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
         }
     }
     u += seg_units;
+    if (u < u_end) {  // re-zero for the next tile segment
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn) acc[fm][fn] = {};
+      if constexpr (ABFT) {
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn) cc[fn] = cw[fn] = 0.f;
+      }
+    }
   }
 }
 

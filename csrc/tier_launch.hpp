@@ -82,7 +82,13 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (mode != 1) {
     const float rounds = ceilf((float)tiles / (float)G0);
     const float waste = (rounds * G0 - tiles) / (rounds * G0);
-    if (waste < 0.15f) {
+    // measured crossover (profiles/sk_probe r2): stream-K pays when the
+    // classic grid wastes a tail round (tiles > G0, bad remainder) or at
+    // deep underfill ABOVE half-fill (3072: 288 tiles vs G0=512 -> +26%);
+    // below G0/2 the per-split combine traffic exceeds what the extra
+    // fill recovers (2560: 200 tiles -> -24%), so small sizes keep the
+    // classic grid (their fix is the finer tiers, not K-splitting).
+    if (waste < 0.15f || 2 * tiles < G0) {
       if (dbg)
         fprintf(stderr, "[sk %dx%d] M=%d N=%d: classic (waste %.3f)\n", BM,
                 BN, M, N, waste);
