@@ -296,35 +296,41 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
     // kernel prologue and keeps them live through the whole hot loop
     // (+68 VGPRs measured on the 64x64-wave tile -> spills on the bigger
     // tiles).
-    int sub_o = sub;
-    asm volatile("" : "+v"(sub_o));
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
-      // fp64 column sums: the fault itself (|e| ~ 1e4) sits in this sum, so
-      // an fp32 accumulation rounds every later term at ulp(1e4) ~ 1e-3 and
-      // the CORRECTION inherits ~5e-3..1e-2 of error (found by
-      // tools/soak.py at |alpha| > 1).  Doubles are fine here: this is the
-      // cold path, entered only by a wave that absorbed a fault.
-      double colp = 0.0, colw = 0.0;
+      // fp64 PLAIN column sum: the fault itself (|e| ~ 1e4) sits in this
+      // sum, so an fp32 accumulation rounds every later term at ulp(1e4)
+      // ~ 1e-3 and the CORRECTION (rc is subtracted from the output)
+      // inherits ~5e-3..1e-2 of error (found by tools/soak.py at
+      // |alpha| > 1).  The WEIGHTED sum stays fp32: rw only selects an
+      // integer row via round(rw/rc), its baseline cw is fp32-maintained
+      // anyway, and the fp32 noise is ~1e-4 of a row index — while the
+      // fp64 weight conversions were what spilled ~143 dwords/lane in the
+      // stream-K fused variants (cold-path scratch traffic ~1 GB/GEMM).
+      int sub_o = sub;  // fresh opaque copy per fn: stops the
+      asm volatile("" : "+v"(sub_o));  // weights being CSE'd+spilled
+      double colp = 0.0;
+      float colwf = 0.f;
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
         for (int reg = 0; reg < NREG; ++reg) {
-          const double v = (double)acc[fm][fn][reg];
-          colp += v;
-          colw = fma((double)(fm * MM + acc_row(reg, sub_o)), v, colw);
+          const float v = acc[fm][fn][reg];
+          colp += (double)v;
+          colwf = fmaf((float)(fm * MM + acc_row(reg, sub_o)), v, colwf);
         }
       const float rc =
           (float)(dslice_sum<MM>(colp) - (double)slice_sum<MM>(cc[fn]));
-      const float rw =
-          (float)(dslice_sum<MM>(colw) - (double)slice_sum<MM>(cw[fn]));
+      const float rw = slice_sum<MM>(colwf) - slice_sum<MM>(cw[fn]);
       const bool cbad = fabsf(rc) > tau;
       const int row = (int)rintf(rw / (cbad ? rc : 1.f));
+      int sub_c = sub;  // independent opaque copy for the correct pass
+      asm volatile("" : "+v"(sub_c));
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
         for (int reg = 0; reg < NREG; ++reg) {
-          const bool hit = cbad && (fm * MM + acc_row(reg, sub_o) == row);
+          const bool hit = cbad && (fm * MM + acc_row(reg, sub_c) == row);
           acc[fm][fn][reg] -= hit ? rc : 0.f;
         }
     }

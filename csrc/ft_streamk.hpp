@@ -155,30 +155,34 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
     // (csrc/ft_kernels.hpp locate_correct/verify_correct), operating on
     // this workgroup's PARTIAL k-range accumulation.
     auto locate_correct = [&]() __attribute__((always_inline)) {
-      int sub_o = sub;
-      asm volatile("" : "+v"(sub_o));
 #pragma unroll
       for (int fn = 0; fn < FN; ++fn) {
-        double colp = 0.0, colw = 0.0;
+        // fp64 plain sum (correction magnitude), fp32 weighted sum (row
+        // index only) — see the classic kernel's locate_correct note.
+        int sub_o = sub;  // fresh opaque copy per fn (see classic)
+        asm volatile("" : "+v"(sub_o));
+        double colp = 0.0;
+        float colwf = 0.f;
 #pragma unroll
         for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
           for (int reg = 0; reg < NREG; ++reg) {
-            const double v = (double)acc[fm][fn][reg];
-            colp += v;
-            colw = fma((double)(fm * MM + acc_row(reg, sub_o)), v, colw);
+            const float v = acc[fm][fn][reg];
+            colp += (double)v;
+            colwf = fmaf((float)(fm * MM + acc_row(reg, sub_o)), v, colwf);
           }
         const float rc =
             (float)(dslice_sum<MM>(colp) - (double)slice_sum<MM>(cc[fn]));
-        const float rw =
-            (float)(dslice_sum<MM>(colw) - (double)slice_sum<MM>(cw[fn]));
+        const float rw = slice_sum<MM>(colwf) - slice_sum<MM>(cw[fn]);
         const bool cbad = fabsf(rc) > tau;
         const int row = (int)rintf(rw / (cbad ? rc : 1.f));
+        int sub_c = sub;  // independent opaque copy for the correct pass
+        asm volatile("" : "+v"(sub_c));
 #pragma unroll
         for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
           for (int reg = 0; reg < NREG; ++reg) {
-            const bool hit = cbad && (fm * MM + acc_row(reg, sub_o) == row);
+            const bool hit = cbad && (fm * MM + acc_row(reg, sub_c) == row);
             acc[fm][fn][reg] -= hit ? rc : 0.f;
           }
       }

@@ -528,13 +528,23 @@ __global__ __launch_bounds__(256, 1) void k_dtvasm2(
 // k_dtvasm16: the D9 asm skeleton with rocBLAS's MFMA choice
 // (v_mfma_f32_16x16x4_f32): fragment rows land on 16-lane groups, so the
 // four fm offsets become immediate `offset:` fields (3 pointer inputs).
-template <int OCC = 2>
+// PAD (D15): pad each B k-row by PAD*16 floats.  Unpadded, the row stride
+// (128 floats) is 0 mod 64 banks, so the four 16-lane s16 groups of a
+// wave hit the SAME 16 banks on every ds_read_b32 of B — a 4-way
+// conflict 8x per k-step.  PAD=1 (stride 144 ≡ 16 mod 64) lands each s16
+// group on its own 16-bank quarter: conflict-free reads.  This is
+// exactly Tensile's LPB16 (LDS pad B 16) in the rocBLAS kernel name —
+// and it is only possible with register->ds_write staging: the glds
+// (HBM->LDS DMA) path of the shipped kernel writes LDS contiguously and
+// cannot pad.
+template <int OCC = 2, int PAD = 0>
 __global__ __launch_bounds__(256, OCC) void k_dtvasm16(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
     float beta) {
   constexpr int BKT = 8;
-  __shared__ __attribute__((aligned(16))) float Bs[2 * 128 * BKT];
+  constexpr int LS = 128 + PAD * 16;  // padded LDS row stride (floats)
+  __shared__ __attribute__((aligned(16))) float Bs[2 * LS * BKT];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int s16 = lane >> 4, r16 = lane & 15;
@@ -581,17 +591,17 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16(
   ISSUE(0, breg, aA);
   WAIT_B(8);
   WAIT_A(0, aA);
-  *(f32x4*)(&Bs[0] + fB) = breg;
+  *(f32x4*)(&Bs[0] + (fB >> 7) * LS + (fB & 127)) = breg;
   __syncthreads();
 
 #define KKLOOP(q, ao)                                                       \
   do {                                                                      \
-    const float* Bp = &Bs[(q) * 128 * BKT];                                 \
+    const float* Bp = &Bs[(q) * LS * BKT];                                 \
     _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                      \
       const int kloc = kk * 4 + s16;                                        \
       float b[8];                                                           \
       _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                      \
-          b[fn] = Bp[kloc * 128 + fn * 16 + r16];                           \
+          b[fn] = Bp[kloc * LS + fn * 16 + r16];                           \
       _Pragma("unroll") for (int fm = 0; fm < 4; ++fm)                      \
           _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                  \
               acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x4f32(           \
@@ -611,7 +621,7 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16(
     KKLOOP(q, acur);                                                        \
     if ((it) + 1 < niter) {                                                 \
       WAIT_B(8);                                                            \
-      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = breg;                      \
+      *(f32x4*)(&Bs[(q ^ 1) * LS * BKT] + (fB >> 7) * LS + (fB & 127)) = breg;                      \
     }                                                                       \
     __syncthreads();                                                        \
   } while (0)
@@ -1001,6 +1011,66 @@ int main(int argc, char** argv) {
       hipEventElapsedTime(&ms, b0, b1);
       printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
              "D12 dtvasm16 16x16x4 asm",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // D15: D12 + LDS-pad-B (Tensile LPB16): conflict-free B reads
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL((k_dtvasm16<2, 1>), grid, block, 0, 0, n, n, n, dA,
+                         dB, dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL((k_dtvasm16<2, 1>), grid, block, 0, 0, n, n, n,
+                           dA, dB, dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL((k_dtvasm16<2, 1>), grid, block, 0, 0, n, n, n,
+                           dA, dB, dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D15 dtvasm16 LPB16 padded",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // depth-2 pipeline on the 16x16x4 asm skeleton (D14)
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm16d2<2>, grid, block, 0, 0, n, n, n, dA, dB,
+                         dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm16d2<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm16d2<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D14 dtvasm16 depth-2",
              2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
              hipGetErrorString(hipGetLastError()));
     }
