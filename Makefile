@@ -15,9 +15,19 @@ KERNEL_TUS := $(filter-out %_hip.hip, $(wildcard csrc/generated/kernel_*.hip))
 
 bin/ft_sgemm: csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip csrc/ft_kernels.hpp csrc/tier_launch.hpp csrc/ft_core.h csrc/generated/tile_params.h $(KERNEL_TUS)
 	mkdir -p bin
-	$(HIPCC) -x hip --offload-arch=$(ARCH) -O3 -std=c++17 -Icsrc \
+	$(HIPCC) -x hip --offload-arch=$(ARCH) -O3 -std=c++17 -Icsrc $(FTFLAGS) \
 	  csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip $(KERNEL_TUS) \
-	  -L$(ROCM)/lib -lrocblas -o $@
+	  -L$(ROCM)/lib -lrocblas -lroctx64 -o $@
+
+# Paranoid-sync build for the race check (tools/race_check.sh): every
+# async-staging site gets an immediate full vmcnt/lgkm drain + barrier;
+# outputs must be BIT-IDENTICAL to the normal build (same per-thread
+# arithmetic order) — any difference indicates a staging race.
+cli-paranoid:
+	mkdir -p bin
+	$(HIPCC) -x hip --offload-arch=$(ARCH) -O3 -std=c++17 -Icsrc -DFT_PARANOID \
+	  csrc/cli_main.hip csrc/dispatch.hip csrc/rocblas_path.hip $(KERNEL_TUS) \
+	  -L$(ROCM)/lib -lrocblas -lroctx64 -o bin/ft_sgemm_paranoid
 
 ext:
 	# ninja does not track header deps for hipcc sources: rebuild .hip TUs

@@ -167,6 +167,17 @@ int main(int argc, char** argv) {
                        hipMemcpyDeviceToHost));
     HIP_CALL(hipMemcpy(hCref.data(), dCref, (size_t)vM * vN * sizeof(float),
                        hipMemcpyDeviceToHost));
+    // FT_SGEMM_DUMP=<dir>: write the raw verification output per kernel id
+    // (bit-exact comparison between the normal and FT_PARANOID builds —
+    // tools/race_check.sh)
+    if (const char* dd = getenv("FT_SGEMM_DUMP")) {
+      char path[512];
+      snprintf(path, sizeof path, "%s/k%02d.bin", dd, kid);
+      if (FILE* df = fopen(path, "wb")) {
+        fwrite(hC.data(), sizeof(float), (size_t)vM * vN, df);
+        fclose(df);
+      }
+    }
     bool ok = true;
     for (size_t i = 0; i < (size_t)vM * vN; ++i) {
       float diff = fabsf(hC[i] - hCref[i]);

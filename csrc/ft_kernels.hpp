@@ -41,6 +41,23 @@
 
 #include <hip/hip_runtime.h>
 
+// FT_PARANOID (race-check build, `make cli-paranoid`): drain ALL
+// outstanding async staging (glds vmcnt + LDS lgkm) and barrier right at
+// every staging site.  Per-thread arithmetic order is unchanged, so the
+// output must be BIT-IDENTICAL to the normal build; any difference is a
+// staging race (SURVEY.md §5 race-detection row).
+#ifdef FT_PARANOID
+#define FT_PARA_SYNC()                                                  \
+  do {                                                                  \
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");         \
+    __syncthreads();                                                    \
+  } while (0)
+#else
+#define FT_PARA_SYNC() \
+  do {                 \
+  } while (0)
+#endif
+
 namespace ftsgemm {
 
 using f32x2 = __attribute__((ext_vector_type(2))) float;
@@ -369,6 +386,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
   constexpr int GLD = GA + GB;  // glds instructions per panel
   stage(0, 0);
   if constexpr (ABFT) strip_stage(0, 0);  // window (panels 0..PPS-1)
+  FT_PARA_SYNC();
   if constexpr (PIPE) {
     // 3-buffer ring: prologue stages two panels; the main loop keeps the
     // newest one in flight across each barrier.
@@ -431,6 +449,7 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
                                                (it + 1) * BK);
         }
       }
+      FT_PARA_SYNC();
       const float* As = &lds[q * BUF];
       const float* Bs = &lds[q * BUF + BM * BK];
       const float* strip =

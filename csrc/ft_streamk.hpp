@@ -217,6 +217,7 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
 
     stage(0, kbase, im0, jn0);
     if constexpr (ABFT) strip_stage(w_lo & 1, kbase, segA);
+    FT_PARA_SYNC();
     __syncthreads();
 
     const int PPG = istride * PPS;  // panels per inject+verify group
@@ -239,6 +240,7 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
                           kbase + (p + 1) * BK, segA);
           }
         }
+        FT_PARA_SYNC();
         const float* As = &lds[qb * BUF];
         const float* Bs = &lds[qb * BUF + BM * BK];
         const float* strip =

@@ -27,6 +27,7 @@
 
 #include <hip/hip_runtime.h>
 #include <rocblas/rocblas.h>
+#include <roctracer/roctx.h>
 
 #include <cstdlib>
 #include <cstring>
@@ -174,9 +175,12 @@ int baseline_ft_sgemm(int M, int N, int K, const float* A, const float* B,
     const float* Bp = B + (size_t)k0 * N;
     const float beta_run = (k0 == 0) ? beta : 1.f;
     // Panel GEMM: C = alpha * Ap * Bp^T + beta_run * C
+    roctxRangePush("baseline_panel_gemm");
     RB_CHECK(rocblas_sgemm(h, rocblas_operation_none,
                            rocblas_operation_transpose, M, N, kp, &alpha, Ap,
                            M, Bp, N, &beta_run, C, M));
+    roctxRangePop();
+    roctxRangePush("baseline_checksum_update");
     // Panel operand sums: s_a = Ap^T e_M, s_b = Bp^T e_N
     if (chain) {
       RB_CHECK(rocblas_sgemv(h, rocblas_operation_transpose, M, kp, &one, Ap,
@@ -194,7 +198,9 @@ int baseline_ft_sgemm(int M, int N, int K, const float* A, const float* B,
                            ws.s_b, 1, &one, ws.ref_row, 1));
     RB_CHECK(rocblas_sgemv(h, rocblas_operation_none, N, kp, &alpha, Bp, N,
                            ws.s_a, 1, &one, ws.ref_col, 1));
+    roctxRangePop();  // baseline_checksum_update
     if ((p + 1) % vevery != 0 && p != npanels - 1) continue;
+    roctxRangePush("baseline_verify_verdict");
     // Observed sums of the running C
     if (chain) {
       RB_CHECK(rocblas_sgemv(h, rocblas_operation_none, M, N, &one, C, M,
@@ -222,6 +228,7 @@ int baseline_ft_sgemm(int M, int N, int K, const float* A, const float* B,
     RB_CHECK(
         rocblas_sdot(h, N, ws.col_c, 1, ws.col_c, 1, ws.d_res + 2 * pv + 1));
     RB_CHECK(rocblas_set_pointer_mode(h, rocblas_pointer_mode_host));
+    roctxRangePop();  // baseline_verify_verdict
     ++pv;
   }
   std::vector<float> res(2 * pv);

@@ -6,6 +6,7 @@
 #pragma once
 
 #include <hip/hip_runtime.h>
+#include <roctracer/roctx.h>
 
 #include <cstdio>
 #include <cstdlib>
@@ -14,6 +15,15 @@
 #include "ft_streamk.hpp"
 
 namespace ftsgemm {
+
+// roctx phase ranges (SURVEY.md §5 tracing row / VERDICT r01 next #9):
+// every launch phase is bracketed so `rocprofv3 --marker-trace
+// --kernel-trace` shows encode / GEMM / verify spans by name.  roctx is a
+// no-op unless a tracer is attached.
+struct RoctxRange {
+  explicit RoctxRange(const char* name) { roctxRangePush(name); }
+  ~RoctxRange() { roctxRangePop(); }
+};
 
 // FT_SGEMM_STREAMK: 0 = never, 1 = force (when the shape divides),
 // unset/other = auto (use stream-K when the classic tile-per-workgroup
@@ -88,7 +98,10 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
     // below G0/2 the per-split combine traffic exceeds what the extra
     // fill recovers (2560: 200 tiles -> -24%), so small sizes keep the
     // classic grid (their fix is the finer tiers, not K-splitting).
-    if (waste < 0.15f || 2 * tiles < G0) {
+    // maxblk > 2 (fine-grained tiers, e.g. large at 10 blocks/CU): the
+    // classic grid self-balances by dynamic dispatch (many short blocks
+    // per CU slot) and measured SK LOSES 15-20% at every size -> classic.
+    if (waste < 0.15f || 2 * tiles < G0 || maxblk > 2) {
       if (dbg)
         fprintf(stderr, "[sk %dx%d] M=%d N=%d: classic (waste %.3f)\n", BM,
                 BN, M, N, waste);
@@ -115,10 +128,14 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (abft) {
     if (!ws) return hipErrorInvalidValue;
     sstr = abft_sstr(K);
+    RoctxRange rr("abft_encode_segsum");
     hipLaunchKernelGGL((segsum_kernel<WM>), dim3(K), dim3(256), 0, stream, M,
                        K, sstr, A, ws);
     SA = ws;
   }
+  RoctxRange rr_main(abft ? (inject ? "ft_sgemm_streamk_abft_inject"
+                                    : "ft_sgemm_streamk_abft")
+                          : "sgemm_streamk_plain");
   // inject+verify cadence: one pass per `istride` strip windows, targeting
   // ~verify_windows groups per tile (reference: 20 per GEMM)
   const int upt = K >> 6;
@@ -196,10 +213,13 @@ hipError_t launch_tier(bool abft, bool inject, int M, int N, int K,
     // replaces — tools/probe_ablate.hip).
     if (!ws) return hipErrorInvalidValue;
     sstr = abft_sstr(K);
+    RoctxRange rr("abft_encode_segsum");
     hipLaunchKernelGGL((segsum_kernel<WM>), dim3(K), dim3(256), 0, stream, M,
                        K, sstr, A, ws);
     SA = ws;
   }
+  RoctxRange rr_main(abft ? (inject ? "ft_sgemm_abft_inject" : "ft_sgemm_abft")
+                          : "sgemm_plain");
   if (abft && inject) {
     hipLaunchKernelGGL((sgemm_mfma<BM, BN, BK, WM, WN, MM, true, true>), grid,
                        block, 0, stream, M, N, K, A, B, C, alpha, beta,

@@ -31,7 +31,7 @@ setup(
             ] + sorted(p for p in glob.glob("csrc/generated/kernel_*.hip")
                        if not p.endswith("_hip.hip")),
             include_dirs=[os.path.join(ROOT, "csrc")],
-            libraries=["rocblas"],
+            libraries=["rocblas", "roctx64"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
