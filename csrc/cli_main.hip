@@ -108,7 +108,9 @@ int main(int argc, char** argv) {
   if (json_env && *json_env) jf = fopen(json_env, "w");
   const int reps = 5;
   const float alpha = 1.f;
-  const size_t maxn = (size_t)end;
+  // END < START runs the verification pass only (tools/race_check.sh);
+  // buffers must still cover the verify size
+  const size_t maxn = (size_t)(end > start ? end : start);
   const size_t bytes = maxn * maxn * sizeof(float);
 
   // Host operands: uniform (-0.9, 0.9), deterministic seed (reference:

@@ -35,11 +35,11 @@ TILING = OrderedDict(
     # ~11% fused at N=4096 (probe_pipeline PROBE_ONLY=T)
     large=dict(bm=64, bn=64, bk=16, wm=32, wn=64, mfma="f32_32x32x2",
                streamk=True),
-    # tall: 16x16x4 MFMA fragments — FM=4 x FN=2 gives 8 MFMAs per k-step
-    # against 6 encode fmas (the 32x32x2 shape had 2 MFMAs vs 2, the worst
-    # encode amortisation of any tier and the one fused-overhead miss vs
-    # the reference in r01: 21% vs 19.9%)
-    tall=dict(bm=128, bn=32, bk=16, wm=64, wn=32, mfma="f32_16x16x4"),
+    # tall: 16x16x4 fragments (8 MFMAs vs 6 encode fmas per k-step) were
+    # measured EQUAL to 32x32x2 (2 vs 2) — plain 109.3 vs 110.3, fused
+    # 86.8 vs 87.0 at N=4096 — so the encode:MFMA issue ratio is NOT what
+    # pins the tall fused overhead at ~21%; the r1-validated shape stays.
+    tall=dict(bm=128, bn=32, bk=16, wm=64, wn=32, mfma="f32_32x32x2"),
     wide=dict(bm=32, bn=128, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
     # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32
     # at N=4096 (tools/probe_pipeline.hip T3/T6 vs PA) — bigger M-tile cuts
