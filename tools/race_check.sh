@@ -43,5 +43,6 @@ if [ "$fail" = 0 ]; then
 else
   echo "race check FAILED" >> "$LOG"
 fi
+rm -rf "$OUT"  # ~150 MB of raw dumps would blow the gpurun_out merge limit
 cat "$LOG"
 exit $fail
