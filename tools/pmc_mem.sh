@@ -1,11 +1,14 @@
 #!/usr/bin/env bash
-# HBM/L2 traffic counters for the tile-choice claim (VERDICT r01 weak #8:
-# "PMC evidence lacks memory-side counters").  TCC has 4 slots and
-# FETCH_SIZE costs 3 / WRITE_SIZE costs 2 (MI355X_MICROARCH.md §rocprofv3
-# PMC slots), so the two run in separate passes.  Counters-only runs — no
-# trace domains mixed in (pool rule).
+# PMC evidence runs (VERDICT r01 weak #8 + next #9).  Counters-only —
+# never combined with trace domains (pool rule).  TCC has 4 slots;
+# FETCH_SIZE costs 3 and WRITE_SIZE 2, so they run in separate passes
+# (MI355X_MICROARCH.md §rocprofv3 PMC slots).
 #
-# Writes gpurun_out/pmc_mem/ per-pass CSVs + gpurun_out/pmc_mem_summary.txt
+# Pass set:
+#   fetch/write x {kid 6 plain huge, kid 16 fused} at N=4096  -> HBM bytes
+#   SQ wait/active x {classic, stream-K} plain huge at N=4608 -> tail-fill
+# Writes a compact gpurun_out/pmc_mem_summary.txt (raw CSVs are deleted:
+# they would blow the gpurun_out merge limit).
 set -e
 cd "$(dirname "$0")/.."
 export TMPDIR=/tmp
@@ -14,35 +17,47 @@ rm -rf "$OUT" && mkdir -p "$OUT"
 SUM=gpurun_out/pmc_mem_summary.txt
 : > "$SUM"
 
-run_pmc() { # name counters kid extra
-  rocprofv3 --pmc $2 -d "$OUT/$1" -o "$1" -- \
-    python tools/profile_one.py --kid $3 --size 4096 --reps 3 $4 \
-    > "$OUT/$1.run.log" 2>&1 || echo "PASS $1 FAILED" >> "$SUM"
+run_pmc() { # name counters kid size env...
+  local name=$1 counters=$2 kid=$3 size=$4; shift 4
+  env "$@" rocprofv3 --pmc $counters --output-format csv -d "$OUT/$name" \
+    -o "$name" -- python tools/profile_one.py --kid "$kid" --size "$size" \
+    --reps 3 > "$OUT/$name.run.log" 2>&1 || echo "PASS $name FAILED" >> "$SUM"
 }
 
-for kid in 6 16; do
-  run_pmc "fetch_k$kid" "FETCH_SIZE" $kid ""
-  run_pmc "write_k$kid" "WRITE_SIZE" $kid ""
-done
+run_pmc fetch_k6   FETCH_SIZE  6 4096
+run_pmc write_k6   WRITE_SIZE  6 4096
+run_pmc fetch_k16  FETCH_SIZE 16 4096
+run_pmc write_k16  WRITE_SIZE 16 4096
+SQC="SQ_WAIT_ANY,SQ_WAIT_INST_ANY,SQ_ACTIVE_INST_ANY,SQ_WAVE_CYCLES"
+run_pmc sq_classic "$SQC" 6 4608 FT_SGEMM_STREAMK=0
+run_pmc sq_streamk "$SQC" 6 4608 FT_SGEMM_STREAMK=1
 
 python3 - "$OUT" >> "$SUM" <<'EOF'
 import csv, glob, sys, os
 out = sys.argv[1]
-print("HBM traffic per GEMM launch at N=4096 (TCC FETCH_SIZE/WRITE_SIZE,")
-print("counter unit = 32B*32ch... reported as raw counter sums; ideal")
-print("A+B operand bytes per huge-tier GEMM = A*(N/128) + B*(M/256)")
-print("= 64MB*32 + 64MB*16 = 3.0 GiB before cache reuse; C rw = 128 MiB)")
-for f in sorted(glob.glob(os.path.join(out, "*", "*counter_collection.csv"))):
-    tag = f.split(os.sep)[-2]
+print("PMC summary (rocprofv3 counter collection, 3 reps + 2 warmups each).")
+print("FETCH/WRITE_SIZE unit = KiB per the gfx94x formula: hbm_bytes ~=")
+print("(FETCH_SIZE + WRITE_SIZE) * 1024 (guide: FETCH_SIZE undercounts a")
+print("wide stream by ~2x).  Ideal huge-tier A+B operand traffic at 4096 =")
+print("A*(N/BN) + B*(M/BM) = 64MiB*32 + 64MiB*16 = 3.0 GiB per GEMM + C rw.")
+print()
+for f in sorted(glob.glob(os.path.join(out, '**', '*.csv'), recursive=True)):
+    tag = os.path.relpath(f, out).split(os.sep)[0]
     agg = {}
     with open(f) as fh:
-        for row in csv.DictReader(fh):
-            name = row.get("Counter_Name") or row.get("counter_name")
-            val = float(row.get("Counter_Value") or row.get("counter_value") or 0)
-            kern = (row.get("Kernel_Name") or row.get("kernel_name") or "")[:40]
-            agg.setdefault((kern, name), 0.0)
-            agg[(kern, name)] += val
+        rd = csv.DictReader(fh)
+        for row in rd:
+            lk = {k.lower(): v for k, v in row.items()}
+            name = lk.get('counter_name') or lk.get('counter-name')
+            val = lk.get('counter_value') or lk.get('counter-value') or 0
+            kern = (lk.get('kernel_name') or lk.get('kernel-name') or '')[:44]
+            if not name:
+                continue
+            key = (kern, name)
+            agg[key] = agg.get(key, 0.0) + float(val)
     for (kern, name), v in sorted(agg.items()):
-        print(f"{tag:14s} {name:12s} {v:16.0f}  {kern}")
+        print(f"{tag:12s} {name:20s} {v:18.0f}  {kern}")
+    print()
 EOF
+rm -rf "$OUT"
 cat "$SUM"
