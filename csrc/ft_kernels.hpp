@@ -223,9 +223,16 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
                       // fault row = round(residual(cw) / residual(cc)))
 
   // ---- async HBM -> LDS staging (global_load_lds, 16 B per lane) ----
+  // staging passes may be fractional in WHOLE WAVES (e.g. the fused tall
+  // twin at BKF=8 stages a 256-float B panel with a 128-thread block:
+  // GB=0 full passes + a 64-lane remainder by wave 0)
   constexpr int GA = (BM * BK) / (THREADS * 4);  // dwordx4 chunks for A
+  constexpr int GAR = (BM * BK / 4) % THREADS;   // remainder lanes
   constexpr int GB = (BN * BK) / (THREADS * 4);
-  static_assert(GA >= 1 && GB >= 1, "tile too small for this thread count");
+  constexpr int GBR = (BN * BK / 4) % THREADS;
+  static_assert(GAR % 64 == 0 && GBR % 64 == 0,
+                "stage remainder must be whole waves");
+  static_assert(GA + GAR > 0 && GB + GBR > 0, "tile too small");
 
   auto stage = [&](int q, int k0) __attribute__((always_inline)) {
     float* dstA = &lds[q * BUF];
@@ -242,6 +249,20 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
                                                         4),
           16, 0, 0);
     }
+    if constexpr (GAR > 0) {
+      if (tid < GAR) {
+        const int f = (GA * THREADS + tid) * 4;
+        const int k = f / BM, i = f % BM;
+        const float* g = A + (im0 + i) + (size_t)(k0 + k) * M;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)g,
+            (__attribute__((address_space(3))) void*)(dstA +
+                                                      (GA * THREADS +
+                                                       wave * 64) *
+                                                          4),
+            16, 0, 0);
+      }
+    }
 #pragma unroll
     for (int t = 0; t < GB; ++t) {
       const int f = (t * THREADS + tid) * 4;
@@ -253,6 +274,20 @@ __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC) void sgemm_mfma(
                                                     (t * THREADS + wave * 64) *
                                                         4),
           16, 0, 0);
+    }
+    if constexpr (GBR > 0) {
+      if (tid < GBR) {
+        const int f = (GB * THREADS + tid) * 4;
+        const int k = f / BN, j = f % BN;
+        const float* g = B + (jn0 + j) + (size_t)(k0 + k) * N;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)g,
+            (__attribute__((address_space(3))) void*)(dstB +
+                                                      (GB * THREADS +
+                                                       wave * 64) *
+                                                          4),
+            16, 0, 0);
+      }
     }
   };
 

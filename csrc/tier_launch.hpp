@@ -51,7 +51,7 @@ inline int device_cu_count() {
 // Stream-K launch of one tier.  Returns hipErrorNotSupported when the
 // classic launch is preferable (heuristic) or the shape does not divide —
 // the caller then falls through to launch_tier.
-template <int BM, int BN, int BK, int WM, int WN, int MM>
+template <int BM, int BN, int BK, int BKF, int WM, int WN, int MM>
 hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
                                  const float* A, const float* B, float* C,
                                  float alpha, float beta, float tau,
@@ -59,16 +59,17 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
                                  hipStream_t stream) {
   const int mode = streamk_env_mode();
   if (mode == 0) return hipErrorNotSupported;
-  if (M % BM || N % BN || K % 64 || K % BK || M % 4 || N % 4)
+  if (M % BM || N % BN || K % 64 || K % (abft ? BKF : BK) || M % 4 ||
+      N % 4)
     return hipErrorNotSupported;
   constexpr int THREADS = 64 * (BM / WM) * (BN / WN);
 
   const void* kfn;
   if (abft && inject)
-    kfn = (const void*)&sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true,
+    kfn = (const void*)&sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true,
                                            true>;
   else if (abft)
-    kfn = (const void*)&sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true,
+    kfn = (const void*)&sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true,
                                            false>;
   else
     kfn = (const void*)&sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, false,
@@ -155,12 +156,12 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (err == hipSuccess) {
     if (abft && inject) {
       hipLaunchKernelGGL(
-          (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true, true>), dim3(G),
+          (sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true, true>), dim3(G),
           dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
           tau, inj_mag, SA, sstr, partials, flags);
     } else if (abft) {
       hipLaunchKernelGGL(
-          (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, true, false>), dim3(G),
+          (sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true, false>), dim3(G),
           dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
           tau, inj_mag, SA, sstr, partials, flags);
     } else {
@@ -189,15 +190,20 @@ size_t abft_workspace_floats_t(int M, int N, int K) {
   return 2 * (size_t)(M / WM) * abft_sstr(K);
 }
 
-template <int BM, int BN, int BK, int WM, int WN, int MM>
+// BKF: K-panel depth of the fused-ABFT twin (may be smaller than the
+// plain BK so the +1KB/wave checksum strips don't push the block over an
+// LDS-occupancy boundary — kernel_table.py bkf note).
+template <int BM, int BN, int BK, int BKF, int WM, int WN, int MM>
 hipError_t launch_tier(bool abft, bool inject, int M, int N, int K,
                        const float* A, const float* B, float* C, float alpha,
                        float beta, float tau, float inj_mag,
                        int verify_windows, float* ws, hipStream_t stream) {
-  if (M % BM || N % BN || K % BK || M % 4 || N % 4) return hipErrorInvalidValue;
+  const int bk_used = abft ? BKF : BK;
+  if (M % BM || N % BN || K % bk_used || M % 4 || N % 4)
+    return hipErrorInvalidValue;
   dim3 grid(M / BM, N / BN);
   dim3 block(64 * (BM / WM) * (BN / WN));
-  const int niter = K / BK;
+  const int niter = K / bk_used;
   // ~20 verify/inject windows per GEMM (reference period K/20,
   // ft_sgemm_huge.cuh:324-327), whole BK panels; plain kernels run one
   // burst.
@@ -221,11 +227,11 @@ hipError_t launch_tier(bool abft, bool inject, int M, int N, int K,
   RoctxRange rr_main(abft ? (inject ? "ft_sgemm_abft_inject" : "ft_sgemm_abft")
                           : "sgemm_plain");
   if (abft && inject) {
-    hipLaunchKernelGGL((sgemm_mfma<BM, BN, BK, WM, WN, MM, true, true>), grid,
+    hipLaunchKernelGGL((sgemm_mfma<BM, BN, BKF, WM, WN, MM, true, true>), grid,
                        block, 0, stream, M, N, K, A, B, C, alpha, beta,
                        stride, stride, tau, inj_mag, SA, sstr);
   } else if (abft) {
-    hipLaunchKernelGGL((sgemm_mfma<BM, BN, BK, WM, WN, MM, true, false>),
+    hipLaunchKernelGGL((sgemm_mfma<BM, BN, BKF, WM, WN, MM, true, false>),
                        grid, block, 0, stream, M, N, K, A, B, C, alpha, beta,
                        stride, stride, tau, inj_mag, SA, sstr);
   } else {

@@ -18,6 +18,12 @@ mr x nr register tile is replaced by per-wave MFMA fragments).
 Fields per tier:
   bm, bn     block (workgroup) output tile
   bk         K-depth of one LDS panel (double buffered)
+  bkf        K-depth for the FUSED-ABFT twin (defaults to bk).  The ABFT
+             checksum strips add 1 KB/wave of LDS; on the skinny tiers
+             whose plain kernel sits exactly at an LDS-occupancy boundary
+             (tall/wide: 20 KB = 8 blocks/CU) that pushes them over a
+             cliff (22 KB -> 7) — a smaller fused panel buys the
+             occupancy back.
   wm, wn     per-wave output sub-tile (waves = (bm/wm)*(bn/wn))
   mfma       'f32_32x32x2' or 'f32_16x16x4' (f32-input MFMA shapes on gfx950)
 """
@@ -33,14 +39,19 @@ TILING = OrderedDict(
     medium=dict(bm=32, bn=32, bk=16, wm=32, wn=32, mfma="f32_32x32x2"),
     # large: 2 waves of 32x64 beat the single 64x64 wave by ~6% plain and
     # ~11% fused at N=4096 (probe_pipeline PROBE_ONLY=T)
-    large=dict(bm=64, bn=64, bk=16, wm=32, wn=64, mfma="f32_32x32x2",
-               streamk=True),
+    # large: bkf=8 (fused LDS 18 KB -> 10 KB: 8 -> 13+ blocks/CU)
+    large=dict(bm=64, bn=64, bk=16, bkf=8, wm=32, wn=64,
+               mfma="f32_32x32x2", streamk=True),
     # tall: 16x16x4 fragments (8 MFMAs vs 6 encode fmas per k-step) were
     # measured EQUAL to 32x32x2 (2 vs 2) — plain 109.3 vs 110.3, fused
     # 86.8 vs 87.0 at N=4096 — so the encode:MFMA issue ratio is NOT what
     # pins the tall fused overhead at ~21%; the r1-validated shape stays.
-    tall=dict(bm=128, bn=32, bk=16, wm=64, wn=32, mfma="f32_32x32x2"),
-    wide=dict(bm=32, bn=128, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
+    # bkf=8: the fused twin's halved panel keeps 20 KB/block -> 8
+    # blocks/CU (see bkf note above).
+    tall=dict(bm=128, bn=32, bk=16, bkf=8, wm=64, wn=32,
+              mfma="f32_32x32x2"),
+    wide=dict(bm=32, bn=128, bk=16, bkf=8, wm=32, wn=64,
+              mfma="f32_32x32x2"),
     # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32
     # at N=4096 (tools/probe_pipeline.hip T3/T6 vs PA) — bigger M-tile cuts
     # total A/B traffic 25% and the 48 KB LDS keeps 2 blocks/CU resident.
