@@ -311,11 +311,20 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
           // one lane spins (256 threads x many owners hammering the same
           // flag line was an L2 atomic storm); the barrier broadcasts the
           // acquire to the workgroup (the acquiring lane's buffer_inv
-          // refreshes the CU-shared L1 all four waves read through)
+          // refreshes the CU-shared L1 all four waves read through).
+          // BOUNDED: a protocol/visibility failure must surface as a
+          // loud verification MISMATCH (poisoned C), never as a device
+          // spin that wedges the box (~2^21 x s_sleep(32) ~ 3.5 s).
           if (tid == 0) {
+            int guard = 0;
             while (__hip_atomic_load(&flags[gc], __ATOMIC_ACQUIRE,
-                                     __HIP_MEMORY_SCOPE_AGENT) == 0u)
+                                     __HIP_MEMORY_SCOPE_AGENT) == 0u) {
               __builtin_amdgcn_s_sleep(32);
+              if (__builtin_expect(++guard > (1 << 21), 0)) {
+                acc[0][0][0] = __builtin_nanf("");  // poison -> loud fail
+                break;
+              }
+            }
           }
           __syncthreads();
           const float* slot = partials + (size_t)gc * (BM * BN) + tid * TPT;
