@@ -199,8 +199,24 @@ int main(int argc, char** argv) {
   printf("\nMatrix Size");
   for (int n = start; n <= end; n += gap) printf("|%8d", n);
   printf("|\n");
+  // FT_SGEMM_SWEEP_IDS="6,16": restrict the sweep to a comma-separated
+  // kernel-id list (benchmarking/bisection aid; default = all 14 rows,
+  // reference parity)
+  bool row_on[17];
+  for (int i = 0; i < 17; ++i) row_on[i] = true;
+  if (const char* ids = getenv("FT_SGEMM_SWEEP_IDS")) {
+    for (int i = 0; i < 17; ++i) row_on[i] = false;
+    const char* p = ids;
+    while (*p) {
+      int v = atoi(p);
+      if (v >= 0 && v <= 16) row_on[v] = true;
+      while (*p && *p != ',') ++p;
+      if (*p == ',') ++p;
+    }
+  }
   for (int idx = 0; idx < 14; ++idx) {
     const int kid = sweep_ids[idx];
+    if (!row_on[kid]) continue;
     printf("%s", kNames[kid]);
     for (int n = start; n <= end; n += gap) {
       const int M = n, N = n, K = n;
