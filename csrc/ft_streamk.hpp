@@ -18,22 +18,25 @@
 // relies on) each processes a CONTIGUOUS unit range, k-fastest: perfect
 // load balance at any size, no tail round.
 //
-// Split-tile combine is the deterministic owner-fixup protocol, not
-// atomics (an atomic f32 combine was measured 3x SLOWER than the classic
-// straggler grid at N=1024 — 16 workgroups hammering the same 128 KB of
-// C serialize in L2):
+// Split-tile combine is a deterministic TWO-KERNEL reduce:
 //   * a tile fully inside one workgroup's range takes the normal
-//     alpha/beta epilogue directly;
-//   * a workgroup whose range STARTS mid-tile (only its first segment
-//     can) writes its raw partial accumulator to a private workspace
-//     slot — coalesced streaming stores, no contention — and
-//     release-stores a per-workgroup flag;
-//   * the tile's OWNER (the workgroup holding the tile's first unit)
-//     acquire-spins on the flags of the (consecutive) contributor
-//     workgroups, adds their partials from the workspace, and runs the
-//     normal alpha/beta epilogue.  Owners only ever wait on HIGHER
-//     workgroup ids, so the dependency graph is acyclic, and summation
-//     order is deterministic.
+//     alpha/beta epilogue directly in the main kernel;
+//   * split-tile segments write their raw partial accumulator to private
+//     workspace slots (head partial -> slot 2g, tail partial -> 2g+1) —
+//     coalesced streaming stores, no contention;
+//   * sk_fixup_kernel (launched after, one workgroup per tile, identical
+//     thread geometry) sums the slots of a split tile in fixed workgroup
+//     order and applies the alpha/beta epilogue; fully-owned tiles exit
+//     in a few instructions.
+// Two rejected designs, both measured: an atomic f32 combine was 3x
+// SLOWER than the classic straggler grid at N=1024 (16 workgroups
+// hammering the same 128 KB of C serialize in L2), and an in-kernel
+// owner-spins-on-contributor-flags protocol deadlocked whenever the
+// dispatcher did not make the whole grid co-resident (fresh-process CLI
+// runs dispatched partially and hung at N=4608; the same binary was fine
+// after prior in-process GPU activity).  The two-kernel form needs no
+// fences, no atomics and no residency assumption: kernel-boundary
+// ordering publishes the partials.
 //
 // ABFT composes cleanly: each workgroup's per-lane column checksums (cc,
 // cw) cover exactly the k-range it accumulated, so the wave-autonomous
@@ -48,6 +51,86 @@
 
 namespace ftsgemm {
 
+// balanced-range inverse map: the workgroup owning work unit u, for a
+// grid of G ranges over `total` units (q = total/G, rem = total%G)
+__device__ inline int sk_wg_of_unit(int u, int q, int rem) {
+  return (u < rem * (q + 1)) ? u / (q + 1)
+                             : rem + (u - rem * (q + 1)) / q;
+}
+
+// Split-tile combine pass: one workgroup per output tile, with the SAME
+// thread geometry as the main kernel so thread t reads back exactly the
+// fragment elements thread t of each contributor wrote.  Fully-owned
+// tiles exit immediately (the main kernel already wrote them).
+template <int BM, int BN, int WM, int WN, int MM>
+static __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN))
+void sk_fixup_kernel(int M, int N, int upt, int G, float alpha, float beta,
+                     float* __restrict__ C,
+                     const float* __restrict__ partials) {
+  using T = mfma_traits<MM>;
+  constexpr int NREG = T::nreg;
+  constexpr int WAVES_N = BN / WN;
+  constexpr int FM = WM / MM, FN = WN / MM;
+  constexpr int TPT = FM * FN * NREG;
+  const int tile = blockIdx.x;
+  const int ntm = M / BM;
+  const int total = ntm * (N / BN) * upt;
+  const int q = total / G, rem = total % G;
+  const int g0 = sk_wg_of_unit(tile * upt, q, rem);
+  const int gl = sk_wg_of_unit(tile * upt + upt - 1, q, rem);
+  if (g0 == gl) return;  // fully owned -> already epilogued
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int sub = lane / MM;
+  const int r = lane % MM;
+  const int wm_idx = wave / WAVES_N, wn_idx = wave % WAVES_N;
+  const int wi0 = wm_idx * WM;
+  const int wj0 = wn_idx * WN;
+  const int bx = tile % ntm, by = tile / ntm;
+  const int im0 = bx * BM, jn0 = by * BN;
+
+  float s[TPT];
+  {  // g0's tail partial (slot 2*g0+1), then every head partial in (g0,gl]
+    const float* slot = partials + (size_t)(2 * g0 + 1) * (BM * BN) +
+                        tid * TPT;
+#pragma unroll
+    for (int e = 0; e < TPT; ++e) s[e] = slot[e];
+  }
+  for (int gc = g0 + 1; gc <= gl; ++gc) {
+    const float* slot = partials + (size_t)(2 * gc) * (BM * BN) + tid * TPT;
+#pragma unroll
+    for (int e4 = 0; e4 < TPT; e4 += 4) {
+      const f32x4 v = *(const f32x4*)(slot + e4);
+#pragma unroll
+      for (int q4 = 0; q4 < 4; ++q4) s[e4 + q4] += v[q4];
+    }
+  }
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      const int j = jn0 + wj0 + fn * MM + r;
+      float* colbase = C + (size_t)j * M + im0 + wi0 + fm * MM;
+#pragma unroll
+      for (int g4 = 0; g4 < NREG / 4; ++g4) {
+        float* p4 = colbase + 4 * sub + 8 * g4;
+        const int e = (fm * FN + fn) * NREG + 4 * g4;
+        f32x4 out;
+        if (beta != 0.f) {
+          const f32x4 prev = *(const f32x4*)p4;
+#pragma unroll
+          for (int q4 = 0; q4 < 4; ++q4)
+            out[q4] = alpha * s[e + q4] + beta * prev[q4];
+        } else {
+#pragma unroll
+          for (int q4 = 0; q4 < 4; ++q4) out[q4] = alpha * s[e + q4];
+        }
+        *(f32x4*)p4 = out;
+      }
+    }
+}
+
 template <int BM, int BN, int BK, int WM, int WN, int MM, bool ABFT,
           bool INJECT, int OCC = 2>
 __global__ __launch_bounds__(64 * (BM / WM) * (BN / WN), OCC)
@@ -55,8 +138,7 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
                         const float* __restrict__ B, float* __restrict__ C,
                         float alpha, float beta, int istride, float tau,
                         float inj_mag, const float* __restrict__ SA, int sstr,
-                        float* __restrict__ partials,
-                        unsigned* __restrict__ flags) {
+                        float* __restrict__ partials) {
   using T = mfma_traits<MM>;
   constexpr int KSTEP = T::kstep;
   constexpr int NREG = T::nreg;
@@ -277,12 +359,24 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
       if constexpr (ABFT) verify_correct();
     }
 
-    // ---- tile contribution (owner-fixup protocol, see header) ----
+    // ---- tile contribution ----
+    // A tile fully inside this range takes the direct alpha/beta epilogue.
+    // Split-tile segments write their raw partial to a private slot (head
+    // partial, w_lo>0 -> slot 2g; tail partial, w_lo==0 && w_hi<upt ->
+    // slot 2g+1); the fixup kernel launched AFTER this one combines them.
+    // No flags, no spin, no co-residency assumption: an earlier protocol
+    // had split-tile owners acquire-spin on contributor flags, which
+    // deadlocked whenever the dispatcher did not make the whole grid
+    // co-resident (reproduced: fresh-process CLI runs dispatched
+    // partially and hung at N=4608; the same binary was fine once the
+    // process had prior GPU activity).  Kernel-boundary ordering makes
+    // the partials visible to the fixup kernel with no fences at all.
     constexpr int TPT = FM * FN * NREG;  // per-thread floats of one tile
-    if (w_lo != 0) {
-      // contributor: only the FIRST segment of a workgroup can start
-      // mid-tile.  Raw partial -> private slot, then release the flag.
-      float* slot = partials + (size_t)g * (BM * BN) + tid * TPT;
+    const bool full_seg = (w_lo == 0) && (w_hi == upt);
+    if (!full_seg) {
+      float* slot = partials +
+                    ((size_t)(2 * g + (w_lo == 0 ? 1 : 0))) * (BM * BN) +
+                    tid * TPT;
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
@@ -294,55 +388,8 @@ void sgemm_mfma_streamk(int M, int N, int K, const float* __restrict__ A,
             for (int q4 = 0; q4 < 4; ++q4) v[q4] = acc[fm][fn][4 * g4 + q4];
             *(f32x4*)(slot + (fm * FN + fn) * NREG + 4 * g4) = v;
           }
-      __threadfence();   // each thread's stores visible agent-wide ...
-      __syncthreads();   // ... before thread 0 publishes for the group
-      if (tid == 0)
-        __hip_atomic_store(&flags[g], 1u, __ATOMIC_RELEASE,
-                           __HIP_MEMORY_SCOPE_AGENT);
     } else {
-      if (w_hi != upt) {
-        // owner of a split tile: fold in every contributor (consecutive
-        // higher workgroup ids whose ranges start inside this tile)
-        const int last_u = tile * upt + upt - 1;
-        int gl;  // workgroup owning the tile's last unit
-        if (last_u < rem * (q + 1)) gl = last_u / (q + 1);
-        else gl = rem + (last_u - rem * (q + 1)) / q;
-        for (int gc = g + 1; gc <= gl; ++gc) {
-          // one lane spins (256 threads x many owners hammering the same
-          // flag line was an L2 atomic storm); the barrier broadcasts the
-          // acquire to the workgroup (the acquiring lane's buffer_inv
-          // refreshes the CU-shared L1 all four waves read through).
-          // BOUNDED: a protocol/visibility failure must surface as a
-          // loud verification MISMATCH (poisoned C), never as a device
-          // spin that wedges the box (~2^21 x s_sleep(32) ~ 3.5 s).
-          if (tid == 0) {
-            int guard = 0;
-            while (__hip_atomic_load(&flags[gc], __ATOMIC_ACQUIRE,
-                                     __HIP_MEMORY_SCOPE_AGENT) == 0u) {
-              __builtin_amdgcn_s_sleep(32);
-              if (__builtin_expect(++guard > (1 << 21), 0)) {
-                acc[0][0][0] = __builtin_nanf("");  // poison -> loud fail
-                break;
-              }
-            }
-          }
-          __syncthreads();
-          const float* slot = partials + (size_t)gc * (BM * BN) + tid * TPT;
-#pragma unroll
-          for (int fm = 0; fm < FM; ++fm)
-#pragma unroll
-            for (int fn = 0; fn < FN; ++fn)
-#pragma unroll
-              for (int g4 = 0; g4 < NREG / 4; ++g4) {
-                const f32x4 v =
-                    *(const f32x4*)(slot + (fm * FN + fn) * NREG + 4 * g4);
-#pragma unroll
-                for (int q4 = 0; q4 < 4; ++q4)
-                  acc[fm][fn][4 * g4 + q4] += v[q4];
-              }
-        }
-      }
-      // normal alpha/beta epilogue (identical to the classic kernel)
+      // direct alpha/beta epilogue (identical to the classic kernel)
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
 #pragma unroll

@@ -143,33 +143,34 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   int istride = upt / (verify_windows > 0 ? verify_windows : 20);
   if (istride < 1) istride = 1;
 
-  // partial-tile slots (one per workgroup) + per-workgroup flags
+  // split-tile partial slots: 2 per workgroup (head + tail partials)
   float* partials = nullptr;
-  unsigned* flags = nullptr;
-  const size_t pbytes = (size_t)G * BM * BN * sizeof(float);
-  if (hipMallocAsync((void**)&partials, pbytes + G * sizeof(unsigned),
-                     stream) != hipSuccess)
+  const size_t pbytes = (size_t)2 * G * BM * BN * sizeof(float);
+  if (hipMallocAsync((void**)&partials, pbytes, stream) != hipSuccess)
     return hipErrorNotSupported;  // nothing mutated yet; classic path
-  flags = (unsigned*)(partials + (size_t)G * BM * BN);
-  hipError_t err = hipMemsetAsync(flags, 0, G * sizeof(unsigned), stream);
 
+  if (abft && inject) {
+    hipLaunchKernelGGL(
+        (sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true, true>), dim3(G),
+        dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
+        tau, inj_mag, SA, sstr, partials);
+  } else if (abft) {
+    hipLaunchKernelGGL(
+        (sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true, false>), dim3(G),
+        dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
+        tau, inj_mag, SA, sstr, partials);
+  } else {
+    hipLaunchKernelGGL(
+        (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, false, false>), dim3(G),
+        dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
+        tau, inj_mag, SA, sstr, partials);
+  }
+  hipError_t err = hipGetLastError();
   if (err == hipSuccess) {
-    if (abft && inject) {
-      hipLaunchKernelGGL(
-          (sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true, true>), dim3(G),
-          dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
-          tau, inj_mag, SA, sstr, partials, flags);
-    } else if (abft) {
-      hipLaunchKernelGGL(
-          (sgemm_mfma_streamk<BM, BN, BKF, WM, WN, MM, true, false>), dim3(G),
-          dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
-          tau, inj_mag, SA, sstr, partials, flags);
-    } else {
-      hipLaunchKernelGGL(
-          (sgemm_mfma_streamk<BM, BN, BK, WM, WN, MM, false, false>), dim3(G),
-          dim3(THREADS), 0, stream, M, N, K, A, B, C, alpha, beta, istride,
-          tau, inj_mag, SA, sstr, partials, flags);
-    }
+    // combine split tiles (fully-owned tiles exit immediately)
+    hipLaunchKernelGGL((sk_fixup_kernel<BM, BN, WM, WN, MM>), dim3(tiles),
+                       dim3(THREADS), 0, stream, M, N, upt, G, alpha, beta,
+                       C, partials);
     err = hipGetLastError();
   }
   (void)hipFreeAsync(partials, stream);
