@@ -49,13 +49,14 @@ TILING = OrderedDict(
     # bkf=8: the fused twin's halved panel keeps 20 KB/block -> 8
     # blocks/CU (see bkf note above).
     # bkf=8 measured: tall fused overhead 21% -> 15-16% (encode cost
-    # 16.5% -> 2.2%: the occupancy cliff was the real tall bottleneck);
-    # wide measured WORSE with bkf=8 (22.0% vs 19.9%: its B-side panel is
-    # the large 128-wide one — halving BK doubles its barrier count for
-    # the same occupancy win), so wide keeps bkf=bk.
+    # 16.5% -> 2.2%: the LDS-occupancy cliff, 22 KB -> 7 blocks/CU, was
+    # the real tall bottleneck, not the encode's issue slots); wide
+    # same-round A/B also favors bkf=8 (noinj 95.4k vs 84.3k GFLOPS at
+    # 4096 — an earlier revert compared against a different box's r1 run)
     tall=dict(bm=128, bn=32, bk=16, bkf=8, wm=64, wn=32,
               mfma="f32_32x32x2"),
-    wide=dict(bm=32, bn=128, bk=16, wm=32, wn=64, mfma="f32_32x32x2"),
+    wide=dict(bm=32, bn=128, bk=16, bkf=8, wm=32, wn=64,
+              mfma="f32_32x32x2"),
     # 256x128 macro-tile, BK=16: measured 135 TF vs 128 TF for 128x128x32
     # at N=4096 (tools/probe_pipeline.hip T3/T6 vs PA) — bigger M-tile cuts
     # total A/B traffic 25% and the 48 KB LDS keeps 2 blocks/CU resident.
