@@ -93,16 +93,18 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (mode != 1) {
     const float rounds = ceilf((float)tiles / (float)G0);
     const float waste = (rounds * G0 - tiles) / (rounds * G0);
-    // measured crossover (profiles/sk_probe r2): stream-K pays when the
-    // classic grid wastes a tail round (tiles > G0, bad remainder) or at
-    // deep underfill ABOVE half-fill (3072: 288 tiles vs G0=512 -> +26%);
-    // below G0/2 the per-split combine traffic exceeds what the extra
-    // fill recovers (2560: 200 tiles -> -24%), so small sizes keep the
-    // classic grid (their fix is the finer tiers, not K-splitting).
-    // maxblk > 2 (fine-grained tiers, e.g. large at 10 blocks/CU): the
-    // classic grid self-balances by dynamic dispatch (many short blocks
-    // per CU slot) and measured SK LOSES 15-20% at every size -> classic.
-    if (waste < 0.15f || 2 * tiles < G0 || maxblk > 2) {
+    // measured crossover (profiles/sk_probe r2, two-kernel fixup):
+    // stream-K wins in TWO regimes of the fat-block tiers —
+    //  * deep underfill, tiles < G0/4 (classic runs at <15% of the block
+    //    slots: 1024 +32%/+99% plain/fused, 1536 +68%/+101%);
+    //  * tail-round waste above half-fill (3072 +28%, 4608..6144 +5-20%).
+    // The middle band keeps the classic grid (2048: SK measured -45%,
+    // 2560 plain -7%).  maxblk > 2 (fine-grained tiers, e.g. large at 10
+    // blocks/CU): dynamic dispatch self-balances the many short blocks
+    // and SK measured 15-20% WORSE at every size -> classic.
+    const bool deep_underfill = 4 * tiles < G0;
+    const bool tail_waste = waste >= 0.15f && 2 * tiles >= G0;
+    if (maxblk > 2 || !(deep_underfill || tail_waste)) {
       if (dbg)
         fprintf(stderr, "[sk %dx%d] M=%d N=%d: classic (waste %.3f)\n", BM,
                 BN, M, N, waste);
