@@ -93,18 +93,19 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (mode != 1) {
     const float rounds = ceilf((float)tiles / (float)G0);
     const float waste = (rounds * G0 - tiles) / (rounds * G0);
-    // measured crossover (profiles/sk_probe r2, two-kernel fixup):
-    // stream-K wins in TWO regimes of the fat-block tiers —
-    //  * deep underfill, tiles < G0/4 (classic runs at <15% of the block
-    //    slots: 1024 +32%/+99% plain/fused, 1536 +68%/+101%);
-    //  * tail-round waste above half-fill (3072 +28%, 4608..6144 +5-20%).
-    // The middle band keeps the classic grid (2048: SK measured -45%,
-    // 2560 plain -7%).  maxblk > 2 (fine-grained tiers, e.g. large at 10
-    // blocks/CU): dynamic dispatch self-balances the many short blocks
-    // and SK measured 15-20% WORSE at every size -> classic.
-    const bool deep_underfill = 4 * tiles < G0;
-    const bool tail_waste = waste >= 0.15f && 2 * tiles >= G0;
-    if (maxblk > 2 || !(deep_underfill || tail_waste)) {
+    // measured crossover (profiles/sk_probe r2, two-kernel fixup), huge
+    // tier, SK vs classic: 1024 +31/+97%, 1536 +76/+98%, 2048 +56/+34%,
+    // 2560 plain -7% / fused +10%, 3072 +28%, 4608..6144 +5..20%,
+    // 4096 & 5632 (waste < 0.15) ~equal-to-slightly-worse.  So: engage
+    // whenever the classic grid wastes >=15% of a dispatch round, except
+    // the plain kernel's dead band around tiles ~ [G/3, G/2) where its
+    // classic 40%-fill run beats the combine cost.  maxblk > 2
+    // (fine-grained tiers, e.g. large at 10 blocks/CU): dynamic dispatch
+    // self-balances the many short blocks and SK measured 15-20% WORSE
+    // at every size -> classic.
+    const bool plain_dead_band =
+        !abft && (3 * tiles > G0) && (2 * tiles < G0);
+    if (maxblk > 2 || waste < 0.15f || plain_dead_band) {
       if (dbg)
         fprintf(stderr, "[sk %dx%d] M=%d N=%d: classic (waste %.3f)\n", BM,
                 BN, M, N, waste);
@@ -145,7 +146,25 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   int istride = upt / (verify_windows > 0 ? verify_windows : 20);
   if (istride < 1) istride = 1;
 
-  // split-tile partial slots: 2 per workgroup (head + tail partials)
+  // split-tile partial slots: 2 per workgroup (head + tail partials).
+  // One-time: raise the default mempool's release threshold so the pool
+  // RETAINS this allocation across stream syncs — with the default
+  // threshold (0) every sync released the 100-200 MB back to the OS and
+  // each launch re-mapped it, which cost more than the whole GEMM at
+  // N<=2048 in sync-per-call usage (CLI sweep: 1024 huge 19.5k steady
+  // -> 8.5k with per-rep remaps).
+  static const bool pool_init = [] {
+    int dev = 0;
+    hipMemPool_t pool = nullptr;
+    if (hipGetDevice(&dev) == hipSuccess &&
+        hipDeviceGetDefaultMemPool(&pool, dev) == hipSuccess) {
+      uint64_t thr = ~0ull;
+      (void)hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold,
+                                   &thr);
+    }
+    return true;
+  }();
+  (void)pool_init;
   float* partials = nullptr;
   const size_t pbytes = (size_t)2 * G * BM * BN * sizeof(float);
   if (hipMallocAsync((void**)&partials, pbytes, stream) != hipSuccess)
