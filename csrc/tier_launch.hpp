@@ -93,19 +93,24 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
   if (mode != 1) {
     const float rounds = ceilf((float)tiles / (float)G0);
     const float waste = (rounds * G0 - tiles) / (rounds * G0);
-    // measured crossover (profiles/sk_probe r2, two-kernel fixup), huge
-    // tier, SK vs classic: 1024 +31/+97%, 1536 +76/+98%, 2048 +56/+34%,
-    // 2560 plain -7% / fused +10%, 3072 +28%, 4608..6144 +5..20%,
-    // 4096 & 5632 (waste < 0.15) ~equal-to-slightly-worse.  So: engage
-    // whenever the classic grid wastes >=15% of a dispatch round, except
-    // the plain kernel's dead band around tiles ~ [G/3, G/2) where its
-    // classic 40%-fill run beats the combine cost.  maxblk > 2
-    // (fine-grained tiers, e.g. large at 10 blocks/CU): dynamic dispatch
-    // self-balances the many short blocks and SK measured 15-20% WORSE
-    // at every size -> classic.
-    const bool plain_dead_band =
-        !abft && (3 * tiles > G0) && (2 * tiles < G0);
-    if (maxblk > 2 || waste < 0.15f || plain_dead_band) {
+    // measured crossover (profiles/sk_probe r2 steady-state + CLI-protocol
+    // sweeps, two-kernel fixup), huge tier.  Two usage regimes matter:
+    // pipelined launches (torch/bench) amortise the per-launch fixup +
+    // workspace costs and SK wins from 1024 up (+31..+98%); the
+    // reference's sync-per-rep CLI protocol pays them serially, where
+    // plain SK only wins in the tail-waste regime (tiles >= G/2, 3072+:
+    // +5..28%) and fused SK wins from 1536 up.  Shipped gate follows the
+    // sync-protocol data (the conservative one; FT_SGEMM_STREAMK=1
+    // forces SK for pipelined callers):
+    //   engage iff waste >= 15%, outside the ~[G/3, G/2) dead band
+    //   (2560: classic 40%-fill beats the combine), and for the plain
+    //   kernel only in the tail regime (2*tiles >= G).
+    // maxblk > 2 (fine-grained tiers, e.g. large at 10 blocks/CU):
+    // dynamic dispatch self-balances the many short blocks and SK
+    // measured 15-20% WORSE at every size -> classic.
+    const bool dead_band = (3 * tiles > G0) && (2 * tiles < G0);
+    if (maxblk > 2 || waste < 0.15f || dead_band ||
+        (!abft && 2 * tiles < G0)) {
       if (dbg)
         fprintf(stderr, "[sk %dx%d] M=%d N=%d: classic (waste %.3f)\n", BM,
                 BN, M, N, waste);
