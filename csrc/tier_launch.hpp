@@ -108,7 +108,9 @@ hipError_t launch_tier_streamk_t(bool abft, bool inject, int M, int N, int K,
     // maxblk > 2 (fine-grained tiers, e.g. large at 10 blocks/CU):
     // dynamic dispatch self-balances the many short blocks and SK
     // measured 15-20% WORSE at every size -> classic.
-    const bool dead_band = (3 * tiles > G0) && (2 * tiles < G0);
+    // dead band tiles in [G/4, G/2): 2560 classic wins outright and 2048
+    // SK is run-to-run bimodal (34k..94k GFLOPS across sweeps) -> classic
+    const bool dead_band = (4 * tiles >= G0) && (2 * tiles < G0);
     if (maxblk > 2 || waste < 0.15f || dead_band ||
         (!abft && 2 * tiles < G0)) {
       if (dbg)
