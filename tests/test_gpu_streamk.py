@@ -105,6 +105,31 @@ def test_streamk_k_not_multiple_of_64_falls_back():
     check(ref, c)
 
 
+@pytest.mark.parametrize("g", [3, 8, 17])
+def test_streamk_multi_segment_workgroups(g):
+    """Workgroup ranges LONGER than one tile (q > upt): each workgroup has
+    a head partial, full middle tiles, and a tail partial — the shape
+    class where the retired owner-spin protocol deadlocked on partial
+    dispatch (N=4608 in the r2 evidence sweep).  A tiny forced G makes
+    every workgroup span multiple tiles deterministically, covering both
+    fixup slots and the g0==gl early-exit."""
+    os.environ["FT_SGEMM_SK_G"] = str(g)
+    try:
+        m, n, k = 1024, 512, 1024   # huge tier: 16 tiles x 16 units
+        a, b, c = ops.make_operands(m, n, k)
+        ref = ops.torch_reference(a, b, c, 1.0, -0.5)
+        ops.sgemm("huge", a, b, c, 1.0, -0.5)
+        torch.cuda.synchronize()
+        check(ref, c)
+        a, b, c = ops.make_operands(m, n, k)
+        ref = ops.torch_reference(a, b, c, 1.0, 0.0)
+        ops.ft_sgemm("huge", a, b, c, 1.0, 0.0, inject=True)
+        torch.cuda.synchronize()
+        check(ref, c)
+    finally:
+        os.environ.pop("FT_SGEMM_SK_G", None)
+
+
 def test_streamk_high_fault_rate():
     """Dense verify windows on split tiles: istride drops to 1 (one
     inject+verify per 64-k window per tile segment)."""
