@@ -23,7 +23,7 @@ column-major MxK matrix is held as a row-major contiguous (K, M) tensor `a`
 with a[k, i] = A[i, k].
 """
 
-__version__ = "0.2.0"
+__version__ = "0.3.0"
 
 from . import utils  # noqa: F401
 from .kernel_table import KERNEL_TABLE, KERNEL_NAMES, TILING  # noqa: F401

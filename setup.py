@@ -19,7 +19,7 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 
 setup(
     name="ft_sgemm_amd",
-    version="0.2.0",
+    version="0.3.0",
     packages=["ft_sgemm_amd"],
     ext_modules=[
         CUDAExtension(
