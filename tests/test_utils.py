@@ -1,3 +1,4 @@
+import os
 import numpy as np
 
 from ft_sgemm_amd.utils import PerfTable, generate_random_matrix, gflops, verify_matrix
@@ -77,3 +78,32 @@ def test_choose_tier():
     # to the rocBLAS path instead of raising
     assert choose_tier(17, 16, 32) is None
     assert choose_tier(100, 100, 100) is None
+
+
+def test_make_tables_from_sweep_log(tmp_path):
+    """The judged-table generator parses a CLI sweep log and derives
+    per-tier overheads + per-column best-of-rows ratios."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    log = os.path.join(root, "profiles", "cli_sweep_r2.log")
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "make_tables.py"), log,
+         str(tmp_path)], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-500:]
+    ovh = open(tmp_path / "overhead_table.txt").read()
+    assert "huge" in ovh and "ref@4096" in ovh
+    ratios = open(tmp_path / "sweep_ratios.txt").read()
+    assert "best plain" in ratios and " 4096 " in ratios
+
+
+def test_compare_reference_runs():
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "compare_reference.py"),
+         os.path.join(root, "profiles", "cli_sweep_r2.log")],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-500:]
+    assert "fused-ABFT overhead" in r.stdout
