@@ -69,6 +69,12 @@ def main():
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
                     help="cpu: CI smoke of the full multi-rank bench "
                          "contract over gloo (tiny size, torch matmul)")
+    ap.add_argument("--graph", action="store_true",
+                    help="capture the step in a hipGraph and replay it "
+                         "(launch-bound small sizes; replicated mode only "
+                         "— the fused path is capture-safe: no syncs, "
+                         "stream-ordered workspace, stream-K bypasses "
+                         "itself under capture)")
     args = ap.parse_args()
 
     cpu_ci = args.device == "cpu"
@@ -124,6 +130,9 @@ def main():
     else:
         step = lambda: ops.rocblas_sgemm(a, b, c, 1.0, -1.5)
 
+    if args.graph and (cpu_ci or args.mode != "replicated"):
+        raise SystemExit("--graph needs --device cuda --mode replicated")
+
     warm0 = time.perf_counter()
     for _ in range(args.warmup):
         step()
@@ -140,6 +149,19 @@ def main():
             if warmup_extra % 50 == 0:  # async launches: bound the overshoot
                 torch.cuda.synchronize()
         torch.cuda.synchronize()
+    if args.graph:
+        # capture once on a side stream, then replace step with the replay
+        g = torch.cuda.CUDAGraph()
+        s = torch.cuda.Stream()
+        with torch.cuda.stream(s):
+            step()
+        torch.cuda.synchronize()
+        with torch.cuda.graph(g):
+            step()
+        step = g.replay
+        g.replay()
+        torch.cuda.synchronize()
+
     if world > 1:
         dist.barrier()
         if not cpu_ci:
@@ -191,6 +213,7 @@ def main():
                 "M": n, "N": n, "K": n,
                 "alpha": 1.0, "beta": -1.5,
                 "inject": inject, "faults_per_gemm": faults,
+                "graph": args.graph,
                 "warmup_extra": warmup_extra,
                 "parallelism": (f"dp{world}" if args.mode == "replicated"
                                 else f"blockrow{world}"),
