@@ -792,6 +792,270 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16d2(
 }
 
 
+// k_dtvasm16ag (D16): the D12 skeleton with a DEPTH-2 VMEM pipeline —
+// panels it+1 AND it+2 stay in flight across the barrier, so the per-panel
+// A-wait (vmcnt(18)) and B-wait (vmcnt(17)) are satisfied ~always and the
+// only hard sync left is the barrier + lgkm drain.  Register budget:
+// D12's 232 + one more A set (8) + one more B staging reg (4) = 244 < 256
+// -> still 2 waves/SIMD (the 512-entry unified file / 8-granule rule,
+// MI355X_MICROARCH.md §register file).  A sets rotate with period 3, B
+// staging with period 2 -> 6-body unrolled loop (requires niter % 6 == 0
+// handled by the guarded while, niter >= 2).
+template <int OCC = 2>
+__global__ __launch_bounds__(256, OCC) void k_dtvasm16ag(
+    int M, int N, int K, const float* __restrict__ A,
+    const float* __restrict__ B, float* __restrict__ C, float alpha,
+    float beta) {
+  constexpr int BKT = 8;
+  __shared__ __attribute__((aligned(16))) float Bs[2 * 128 * BKT];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int s16 = lane >> 4, r16 = lane & 15;
+  const int wi0 = wave * 64;
+  const int im0 = blockIdx.x * 256, jn0 = blockIdx.y * 128;
+
+  f32x4 acc[4][8] = {};
+  float aA[8], aB[8], aC[8];
+  f32x4 br0, br1;
+
+  const int fB = tid * 4;
+  const float* pB0 = B + (jn0 + (fB & 127)) + (size_t)(fB >> 7) * N;
+  const float* pA0 = A + (im0 + wi0 + r16) + (size_t)s16 * M;
+
+#define ISSUE(p, b0v, ao)                                                   \
+  do {                                                                      \
+    const size_t ko = (size_t)(p) * BKT;                                    \
+    const float* pb = pB0 + ko * N;                                         \
+    const float* q0 = pA0 + ko * M;                                         \
+    const float* q1 = q0 + 4 * (size_t)M;                                   \
+    asm volatile("global_load_dwordx4 %0, %9, off\n\t"                      \
+                 "global_load_dword %1, %10, off\n\t"                       \
+                 "global_load_dword %2, %10, off offset:64\n\t"             \
+                 "global_load_dword %3, %10, off offset:128\n\t"            \
+                 "global_load_dword %4, %10, off offset:192\n\t"            \
+                 "global_load_dword %5, %11, off\n\t"                       \
+                 "global_load_dword %6, %11, off offset:64\n\t"             \
+                 "global_load_dword %7, %11, off offset:128\n\t"            \
+                 "global_load_dword %8, %11, off offset:192"                \
+                 : "=&v"(b0v), "=&v"((ao)[0]), "=&v"((ao)[1]),              \
+                   "=&v"((ao)[2]), "=&v"((ao)[3]), "=&v"((ao)[4]),          \
+                   "=&v"((ao)[5]), "=&v"((ao)[6]), "=&v"((ao)[7])           \
+                 : "v"(pb), "v"(q0), "v"(q1));                              \
+  } while (0)
+#define WAIT_A(n, ao)                                                       \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1"                         \
+               : "+v"((ao)[0]), "+v"((ao)[1]), "+v"((ao)[2]),               \
+                 "+v"((ao)[3]), "+v"((ao)[4]), "+v"((ao)[5]),               \
+                 "+v"((ao)[6]), "+v"((ao)[7]))
+#define WAIT_B(n, bv) \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1" : "+v"(bv))
+
+  const int niter = K / BKT;
+  ISSUE(0, br0, aA);
+  ISSUE(1, br1, aB);
+  WAIT_B(17, br0);  // B0 retired (A0 x8 + batch1 x9 issued after it)
+  *(f32x4*)(&Bs[0] + fB) = br0;
+  __syncthreads();
+
+#define KKLOOP(q, ao)                                                       \
+  do {                                                                      \
+    const float* Bp = &Bs[(q) * 128 * BKT];                                 \
+    _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                      \
+      const int kloc = kk * 4 + s16;                                        \
+      float b[8];                                                           \
+      _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                      \
+          b[fn] = Bp[kloc * 128 + fn * 16 + r16];                           \
+      _Pragma("unroll") for (int fm = 0; fm < 4; ++fm)                      \
+          _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                  \
+              asm volatile("v_mfma_f32_16x16x4_f32 %0, %1, %2, %0"          \
+                           : "+a"(acc[fm][fn])                              \
+                           : "v"((ao)[kk * 4 + fm]), "v"(b[fn]));           \
+    }                                                                       \
+  } while (0)
+
+  // body it: Bs[q] holds B_it (published last body); Acur = A_it (in
+  // flight from 2 bodies ago); BRwr = B_{it+1} staging reg; issue panel
+  // it+2 into (BRiss, Anxt).  Publish B_{it+1} BEFORE the MFMA loop so the
+  // lgkm drain overlaps the MFMA stream instead of stalling the barrier.
+#define BODY(it, Acur, Anxt, BRwr, BRiss)                                   \
+  do {                                                                      \
+    const int q = (it) & 1;                                                 \
+    if ((it) + 2 < niter) {                                                 \
+      ISSUE((it) + 2, BRiss, Anxt);                                         \
+      WAIT_B(17, BRwr);                                                     \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRwr;                      \
+      WAIT_A(18, Acur);                                                     \
+    } else if ((it) + 1 < niter) {                                          \
+      WAIT_B(8, BRwr);                                                      \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRwr;                      \
+      WAIT_A(9, Acur);                                                      \
+    } else {                                                                \
+      WAIT_A(0, Acur);                                                      \
+    }                                                                       \
+    KKLOOP(q, Acur);                                                        \
+    __syncthreads();                                                        \
+  } while (0)
+
+  int it = 0;
+  while (it < niter) {
+    BODY(it, aA, aC, br1, br0);
+    if (++it >= niter) break;
+    BODY(it, aB, aA, br0, br1);
+    if (++it >= niter) break;
+    BODY(it, aC, aB, br1, br0);
+    if (++it >= niter) break;
+    BODY(it, aA, aC, br0, br1);
+    if (++it >= niter) break;
+    BODY(it, aB, aA, br1, br0);
+    if (++it >= niter) break;
+    BODY(it, aC, aB, br0, br1);
+    ++it;
+  }
+#undef ISSUE
+#undef WAIT_A
+#undef WAIT_B
+#undef KKLOOP
+#undef BODY
+
+  asm volatile("s_nop 7" ::: "memory");  // MFMA D -> VALU-reader hazard
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 8; ++fn) {
+      const int j = jn0 + fn * 16 + r16;
+      float* p = C + (size_t)j * M + im0 + wi0 + fm * 16 + s16 * 4;
+      const f32x4 prev = *(const f32x4*)p;
+      f32x4 out;
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        out[u] = alpha * acc[fm][fn][u] + beta * prev[u];
+      *(f32x4*)p = out;
+    }
+}
+
+
+
+// k_dtvasm16p2 (D17): depth-2 A-slack at D12's register budget.  Instead
+// of D14/D16's 3-set rotation (which the allocator shredded: 2.4-2.8 KB
+// scratch), panel it+2 is issued at BODY END into the registers the body
+// just freed (Acur consumed by the MFMA loop, BRwr published to LDS) —
+// period-2 rotation, 2 A-sets + 2 B-regs.  The A-operand wait slack grows
+// from 1 body (D12: ISSUE(it+1) then wait A_it) to 2 bodies, and the only
+// in-body wait left is vmcnt(8) for the B publish (1 body of slack).
+template <int OCC = 2>
+__global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
+    int M, int N, int K, const float* __restrict__ A,
+    const float* __restrict__ B, float* __restrict__ C, float alpha,
+    float beta) {
+  constexpr int BKT = 8;
+  __shared__ __attribute__((aligned(16))) float Bs[2 * 128 * BKT];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int s16 = lane >> 4, r16 = lane & 15;
+  const int wi0 = wave * 64;
+  const int im0 = blockIdx.x * 256, jn0 = blockIdx.y * 128;
+
+  f32x4 acc[4][8] = {};
+  float aA[8], aB[8];
+  f32x4 br0, br1;
+
+  const int fB = tid * 4;
+  const float* pB0 = B + (jn0 + (fB & 127)) + (size_t)(fB >> 7) * N;
+  const float* pA0 = A + (im0 + wi0 + r16) + (size_t)s16 * M;
+
+#define ISSUE(p, b0v, ao)                                                   \
+  do {                                                                      \
+    const size_t ko = (size_t)(p) * BKT;                                    \
+    const float* pb = pB0 + ko * N;                                         \
+    const float* q0 = pA0 + ko * M;                                         \
+    const float* q1 = q0 + 4 * (size_t)M;                                   \
+    asm volatile("global_load_dwordx4 %0, %9, off\n\t"                      \
+                 "global_load_dword %1, %10, off\n\t"                       \
+                 "global_load_dword %2, %10, off offset:64\n\t"             \
+                 "global_load_dword %3, %10, off offset:128\n\t"            \
+                 "global_load_dword %4, %10, off offset:192\n\t"            \
+                 "global_load_dword %5, %11, off\n\t"                       \
+                 "global_load_dword %6, %11, off offset:64\n\t"             \
+                 "global_load_dword %7, %11, off offset:128\n\t"            \
+                 "global_load_dword %8, %11, off offset:192"                \
+                 : "=&v"(b0v), "=&v"((ao)[0]), "=&v"((ao)[1]),              \
+                   "=&v"((ao)[2]), "=&v"((ao)[3]), "=&v"((ao)[4]),          \
+                   "=&v"((ao)[5]), "=&v"((ao)[6]), "=&v"((ao)[7])           \
+                 : "v"(pb), "v"(q0), "v"(q1));                              \
+  } while (0)
+#define WAIT_A(n, ao)                                                       \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1"                         \
+               : "+v"((ao)[0]), "+v"((ao)[1]), "+v"((ao)[2]),               \
+                 "+v"((ao)[3]), "+v"((ao)[4]), "+v"((ao)[5]),               \
+                 "+v"((ao)[6]), "+v"((ao)[7]))
+#define WAIT_B(n, bv) \
+  asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1" : "+v"(bv))
+
+  const int niter = K / BKT;
+  ISSUE(0, br0, aA);
+  if (1 < niter) ISSUE(1, br1, aB);
+  WAIT_B(17, br0);  // B0 retired (A0 x8 + batch1 x9 after it)
+  *(f32x4*)(&Bs[0] + fB) = br0;
+  __syncthreads();
+
+#define KKLOOP(q, ao)                                                       \
+  do {                                                                      \
+    const float* Bp = &Bs[(q) * 128 * BKT];                                 \
+    _Pragma("unroll") for (int kk = 0; kk < 2; ++kk) {                      \
+      const int kloc = kk * 4 + s16;                                        \
+      float b[8];                                                           \
+      _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                      \
+          b[fn] = Bp[kloc * 128 + fn * 16 + r16];                           \
+      _Pragma("unroll") for (int fm = 0; fm < 4; ++fm)                      \
+          _Pragma("unroll") for (int fn = 0; fn < 8; ++fn)                  \
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x4f32(           \
+                  (ao)[kk * 4 + fm], b[fn], acc[fm][fn], 0, 0, 0);          \
+    }                                                                       \
+  } while (0)
+
+  // body it: Bs[it&1] holds B_it (published last body); Acur = A_it (in
+  // flight since body it-2 -> arrived); BRwr = B_{it+1} (issued last
+  // body, 1-body slack).  Publish B_{it+1} BEFORE the MFMA loop; issue
+  // panel it+2 AFTER it, into the registers this body just freed.
+#define BODY(it, Acur, BRwr)                                                \
+  do {                                                                      \
+    const int q = (it) & 1;                                                 \
+    if ((it) + 1 < niter) {                                                 \
+      WAIT_B(8, BRwr);                                                      \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRwr;                      \
+    } else {                                                                \
+      WAIT_A(0, Acur);                                                      \
+    }                                                                       \
+    KKLOOP(q, Acur);                                                        \
+    if ((it) + 2 < niter) ISSUE((it) + 2, BRwr, Acur);                      \
+    __syncthreads();                                                        \
+  } while (0)
+
+  for (int it = 0; it < niter; it += 2) {
+    BODY(it, aA, br1);
+    if (it + 1 < niter) BODY(it + 1, aB, br0);
+  }
+#undef ISSUE
+#undef WAIT_A
+#undef WAIT_B
+#undef KKLOOP
+#undef BODY
+
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 8; ++fn) {
+      const int j = jn0 + fn * 16 + r16;
+      float* p = C + (size_t)j * M + im0 + wi0 + fm * 16 + s16 * 4;
+      const f32x4 prev = *(const f32x4*)p;
+      f32x4 out;
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        out[u] = alpha * acc[fm][fn][u] + beta * prev[u];
+      *(f32x4*)p = out;
+    }
+}
+
 __global__ void fill_lcg(float* p, size_t n, unsigned seed) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
@@ -1071,6 +1335,66 @@ int main(int argc, char** argv) {
       hipEventElapsedTime(&ms, b0, b1);
       printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
              "D14 dtvasm16 depth-2",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // D17: period-2 depth-2 (issue at body end into freed regs)
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm16p2<2>, grid, block, 0, 0, n, n, n, dA,
+                         dB, dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm16p2<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm16p2<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D17 dtvasm16 p2 depth-2",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // D16: depth-2 + AGPR accumulator
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL(k_dtvasm16ag<2>, grid, block, 0, 0, n, n, n, dA,
+                         dB, dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL(k_dtvasm16ag<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL(k_dtvasm16ag<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D16 dtvasm16 AGPR depth-2",
              2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
              hipGetErrorString(hipGetLastError()));
     }
