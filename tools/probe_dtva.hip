@@ -935,13 +935,15 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16ag(
 
 
 
-// k_dtvasm16p2 (D17): depth-2 A-slack at D12's register budget.  Instead
-// of D14/D16's 3-set rotation (which the allocator shredded: 2.4-2.8 KB
-// scratch), panel it+2 is issued at BODY END into the registers the body
-// just freed (Acur consumed by the MFMA loop, BRwr published to LDS) —
-// period-2 rotation, 2 A-sets + 2 B-regs.  The A-operand wait slack grows
-// from 1 body (D12: ISSUE(it+1) then wait A_it) to 2 bodies, and the only
-// in-body wait left is vmcnt(8) for the B publish (1 body of slack).
+// k_dtvasm16p2 (D17b): depth-2 at D12's register budget via SPLIT issues,
+// all period-2 (the first D17 cut reused the published B register for the
+// next issue, which collapsed the rotation to one register and produced
+// wrong panels; D14/D16's 3-set rotation spilled 2.4-2.8 KB/lane).
+//   top(it):  ISSUE_B(it+2) -> br[it&1]   (B_it published last body ->
+//             register free; B wait slack = 1 full body)
+//   wait vmcnt(9): retires B_{it+1} (and everything older, incl. A_it
+//             which has 2 bodies of slack); publish B_{it+1} to LDS
+//   KKLOOP(A_it); ISSUE_A(it+2) -> a[it&1] (just consumed); barrier.
 template <int OCC = 2>
 __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
     int M, int N, int K, const float* __restrict__ A,
@@ -991,10 +993,39 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
 #define WAIT_B(n, bv) \
   asm volatile("s_waitcnt vmcnt(" #n ")\n\ts_nop 1" : "+v"(bv))
 
+#define ISSUE_B(p, b0v)                                                     \
+  do {                                                                      \
+    const float* pb = pB0 + (size_t)(p) * BKT * N;                          \
+    asm volatile("global_load_dwordx4 %0, %1, off"                          \
+                 : "=&v"(b0v) : "v"(pb));                                   \
+  } while (0)
+#define ISSUE_A(p, ao)                                                      \
+  do {                                                                      \
+    const size_t ko = (size_t)(p) * BKT;                                    \
+    const float* q0 = pA0 + ko * M;                                         \
+    const float* q1 = q0 + 4 * (size_t)M;                                   \
+    asm volatile("global_load_dword %0, %8, off\n\t"                        \
+                 "global_load_dword %1, %8, off offset:64\n\t"              \
+                 "global_load_dword %2, %8, off offset:128\n\t"             \
+                 "global_load_dword %3, %8, off offset:192\n\t"             \
+                 "global_load_dword %4, %9, off\n\t"                        \
+                 "global_load_dword %5, %9, off offset:64\n\t"              \
+                 "global_load_dword %6, %9, off offset:128\n\t"             \
+                 "global_load_dword %7, %9, off offset:192"                 \
+                 : "=&v"((ao)[0]), "=&v"((ao)[1]), "=&v"((ao)[2]),          \
+                   "=&v"((ao)[3]), "=&v"((ao)[4]), "=&v"((ao)[5]),          \
+                   "=&v"((ao)[6]), "=&v"((ao)[7])                           \
+                 : "v"(q0), "v"(q1));                                       \
+  } while (0)
+
   const int niter = K / BKT;
-  ISSUE(0, br0, aA);
-  if (1 < niter) ISSUE(1, br1, aB);
-  WAIT_B(17, br0);  // B0 retired (A0 x8 + batch1 x9 after it)
+  ISSUE_B(0, br0);
+  ISSUE_A(0, aA);
+  if (1 < niter) {
+    ISSUE_B(1, br1);
+    ISSUE_A(1, aB);
+  }
+  WAIT_B(17, br0);  // B0 retired (A0 x8 + B1 + A1 x8 = 17 after it)
   *(f32x4*)(&Bs[0] + fB) = br0;
   __syncthreads();
 
@@ -1017,25 +1048,30 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
   // flight since body it-2 -> arrived); BRwr = B_{it+1} (issued last
   // body, 1-body slack).  Publish B_{it+1} BEFORE the MFMA loop; issue
   // panel it+2 AFTER it, into the registers this body just freed.
-#define BODY(it, Acur, BRwr)                                                \
+#define BODY(it, Acur, BRcur, BRnxt)                                        \
   do {                                                                      \
     const int q = (it) & 1;                                                 \
-    if ((it) + 1 < niter) {                                                 \
-      WAIT_B(8, BRwr);                                                      \
-      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRwr;                      \
+    if ((it) + 2 < niter) {                                                 \
+      ISSUE_B((it) + 2, BRcur);                                             \
+      WAIT_B(9, BRnxt);                                                     \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRnxt;                     \
+    } else if ((it) + 1 < niter) {                                          \
+      WAIT_B(8, BRnxt);                                                     \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRnxt;                     \
     } else {                                                                \
       WAIT_A(0, Acur);                                                      \
     }                                                                       \
     KKLOOP(q, Acur);                                                        \
-    if ((it) + 2 < niter) ISSUE((it) + 2, BRwr, Acur);                      \
+    if ((it) + 2 < niter) ISSUE_A((it) + 2, Acur);                          \
     __syncthreads();                                                        \
   } while (0)
 
   for (int it = 0; it < niter; it += 2) {
-    BODY(it, aA, br1);
-    if (it + 1 < niter) BODY(it + 1, aB, br0);
+    BODY(it, aA, br0, br1);
+    if (it + 1 < niter) BODY(it + 1, aB, br1, br0);
   }
-#undef ISSUE
+#undef ISSUE_B
+#undef ISSUE_A
 #undef WAIT_A
 #undef WAIT_B
 #undef KKLOOP
