@@ -944,7 +944,7 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16ag(
 //   wait vmcnt(9): retires B_{it+1} (and everything older, incl. A_it
 //             which has 2 bodies of slack); publish B_{it+1} to LDS
 //   KKLOOP(A_it); ISSUE_A(it+2) -> a[it&1] (just consumed); barrier.
-template <int OCC = 2>
+template <int OCC = 2, bool SAFE = false>
 __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
     int M, int N, int K, const float* __restrict__ A,
     const float* __restrict__ B, float* __restrict__ C, float alpha,
@@ -1053,15 +1053,18 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
     const int q = (it) & 1;                                                 \
     if ((it) + 2 < niter) {                                                 \
       ISSUE_B((it) + 2, BRcur);                                             \
-      WAIT_B(9, BRnxt);                                                     \
-      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRnxt;                     \
+      WAIT_A(10, Acur);                                                     \
     } else if ((it) + 1 < niter) {                                          \
-      WAIT_B(8, BRnxt);                                                     \
-      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRnxt;                     \
+      WAIT_A(9, Acur);                                                      \
     } else {                                                                \
       WAIT_A(0, Acur);                                                      \
     }                                                                       \
     KKLOOP(q, Acur);                                                        \
+    if ((it) + 1 < niter) {                                                 \
+      if constexpr (SAFE) WAIT_B(2, BRnxt);                                 \
+      else WAIT_B(6, BRnxt);                                                \
+      *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRnxt;                     \
+    }                                                                       \
     if ((it) + 2 < niter) ISSUE_A((it) + 2, Acur);                          \
     __syncthreads();                                                        \
   } while (0)
@@ -1098,6 +1101,18 @@ __global__ void fill_lcg(float* p, size_t n, unsigned seed) {
   unsigned s = seed ^ (unsigned)(i * 2654435761u);
   s ^= s << 13; s ^= s >> 17; s ^= s << 5;
   p[i] = ((s >> 8) * (1.0f / 16777216.0f)) * 1.8f - 0.9f;
+}
+
+// bad-element census for debugging: counts |diff|>0.1 and records the
+// first few linear indices (column-major C: idx = i + j*n)
+__global__ void bad_census(const float* x, const float* y, size_t n,
+                           unsigned* count, unsigned* idxs) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  if (fabsf(x[i] - y[i]) > 0.1f) {
+    unsigned slot = atomicAdd(count, 1u);
+    if (slot < 16) idxs[slot] = (unsigned)i;
+  }
 }
 
 __global__ void max_diff(const float* x, const float* y, size_t n,
@@ -1399,8 +1414,57 @@ int main(int argc, char** argv) {
       hipEventSynchronize(b1);
       float ms;
       hipEventElapsedTime(&ms, b0, b1);
+      {
+        unsigned *dCnt;
+        hipMalloc(&dCnt, 4 + 16 * 4);
+        hipMemsetAsync(dCnt, 0, 4 + 16 * 4, 0);
+        hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+        hipLaunchKernelGGL(k_dtvasm16p2<2>, grid, block, 0, 0, n, n, n, dA,
+                           dB, dC, 1.f, 0.f);
+        hipLaunchKernelGGL(bad_census, dim3(((size_t)n * n + 255) / 256),
+                           dim3(256), 0, 0, dC, dRef, (size_t)n * n, dCnt,
+                           dCnt + 1);
+        unsigned h[17];
+        hipMemcpy(h, dCnt, sizeof h, hipMemcpyDeviceToHost);
+        printf("D17 bad elements: %u of %zu; first idx (i,j): ", h[0],
+               (size_t)n * n);
+        for (int u = 0; u < 8 && u < (int)h[0]; ++u)
+          printf("(%u,%u) ", h[1 + u] % n, h[1 + u] / n);
+        printf("\n");
+        hipFree(dCnt);
+      }
       printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
              "D17 dtvasm16 p2 depth-2",
+             2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
+             hipGetErrorString(hipGetLastError()));
+    }
+    {  // D17-SAFE: same rotation, full-drain waits (wait-count bisect)
+      dim3 grid(n / BM, n / BN), block(256);
+      hipMemsetD32Async((hipDeviceptr_t)dC, 0, (size_t)n * n, 0);
+      hipLaunchKernelGGL((k_dtvasm16p2<2, true>), grid, block, 0, 0, n, n,
+                         n, dA, dB, dC, 1.f, 0.f);
+      float md = 0.f;
+      hipMemcpyAsync(dMax, &md, 4, hipMemcpyHostToDevice, 0);
+      hipLaunchKernelGGL(max_diff, dim3(((size_t)n * n + 255) / 256),
+                         dim3(256), 0, 0, dC, dRef, (size_t)n * n, dMax);
+      hipMemcpy(&md, dMax, 4, hipMemcpyDeviceToHost);
+      hipEvent_t b0, b1;
+      hipEventCreate(&b0);
+      hipEventCreate(&b1);
+      for (int w = 0; w < 2; ++w)
+        hipLaunchKernelGGL((k_dtvasm16p2<2, true>), grid, block, 0, 0, n,
+                           n, n, dA, dB, dC, 1.f, -1.5f);
+      hipDeviceSynchronize();
+      hipEventRecord(b0);
+      for (int rr = 0; rr < reps; ++rr)
+        hipLaunchKernelGGL((k_dtvasm16p2<2, true>), grid, block, 0, 0, n,
+                           n, n, dA, dB, dC, 1.f, -1.5f);
+      hipEventRecord(b1);
+      hipEventSynchronize(b1);
+      float ms;
+      hipEventElapsedTime(&ms, b0, b1);
+      printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
+             "D17S dtvasm16 p2 SAFE-waits",
              2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
              hipGetErrorString(hipGetLastError()));
     }
