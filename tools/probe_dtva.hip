@@ -1053,7 +1053,8 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
     const int q = (it) & 1;                                                 \
     if ((it) + 2 < niter) {                                                 \
       ISSUE_B((it) + 2, BRcur);                                             \
-      WAIT_A(10, Acur);                                                     \
+      if constexpr (SAFE) WAIT_A(0, Acur);                                  \
+      else WAIT_A(10, Acur);                                                \
     } else if ((it) + 1 < niter) {                                          \
       WAIT_A(9, Acur);                                                      \
     } else {                                                                \
@@ -1061,8 +1062,8 @@ __global__ __launch_bounds__(256, OCC) void k_dtvasm16p2(
     }                                                                       \
     KKLOOP(q, Acur);                                                        \
     if ((it) + 1 < niter) {                                                 \
-      if constexpr (SAFE) WAIT_B(2, BRnxt);                                 \
-      else WAIT_B(6, BRnxt);                                                \
+      if constexpr (SAFE) WAIT_B(0, BRnxt);                                 \
+      else WAIT_B(2, BRnxt);                                                \
       *(f32x4*)(&Bs[(q ^ 1) * 128 * BKT] + fB) = BRnxt;                     \
     }                                                                       \
     if ((it) + 2 < niter) ISSUE_A((it) + 2, Acur);                          \
@@ -1494,7 +1495,7 @@ int main(int argc, char** argv) {
       float ms;
       hipEventElapsedTime(&ms, b0, b1);
       printf("N=%d %-30s %8.0f GFLOPS (maxdiff %.2e) err=%s\n", n,
-             "D16 dtvasm16 AGPR depth-2",
+             "D16 AGPR depth-2 (spills, see ABLATION)",
              2.0 * n * n * n * reps / (ms * 1e-3) / 1e9, md,
              hipGetErrorString(hipGetLastError()));
     }
