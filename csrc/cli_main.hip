@@ -80,7 +80,9 @@ bool run_kernel(int kid, int M, int N, int K, const float* dA,
     // dominate; FT_SGEMM_PANEL_K overrides)
     static const int panel_k = [] {
       const char* e = getenv("FT_SGEMM_PANEL_K");
-      return e ? atoi(e) : 1024;
+      int p = e ? atoi(e) : 1024;
+      // the verdict-slot buffer is sized for panel_k >= 64 (2*(maxn/64+2))
+      return p < 64 ? 64 : p;
     }();
     float r0 = 0, r1 = 0;
     return ftsgemm::baseline_ft_sgemm(M, N, K, dA, dB, dC, alpha, beta, ws,
