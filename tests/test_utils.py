@@ -107,3 +107,67 @@ def test_compare_reference_runs():
         capture_output=True, text=True)
     assert r.returncode == 0, r.stderr[-500:]
     assert "fused-ABFT overhead" in r.stdout
+
+
+def test_streamk_partition_properties():
+    """Property test of the stream-K work partition + fixup mapping
+    (csrc/ft_streamk.hpp): balanced contiguous ranges cover [0, total)
+    exactly once; for every SPLIT tile, g0's tail-partial slot plus the
+    head-partial slots of (g0, gl] reconstruct exactly the tile's units;
+    fully-owned tiles have no slot writers; no workgroup writes more than
+    one head and one tail slot."""
+    import random
+    rng = random.Random(20260914)
+
+    def wg_of(u, q, r):
+        return u // (q + 1) if u < r * (q + 1) else r + (u - r * (q + 1)) // q
+
+    for _ in range(300):
+        ntiles = rng.randint(1, 80)
+        upt = rng.choice([1, 2, 3, 8, 16, 48, 72])
+        total = ntiles * upt
+        G = rng.randint(1, 64)
+        G = min(G, total)
+        q, r = divmod(total, G)
+
+        # walk each workgroup's segments exactly as the kernel does
+        covered = [0] * total
+        head_writers, tail_writers = {}, {}   # g -> (tile, units)
+        for g in range(G):
+            u = g * q + min(g, r)
+            u_end = u + q + (1 if g < r else 0)
+            first = True
+            while u < u_end:
+                tile = u // upt
+                w_lo = u - tile * upt
+                seg = min(u_end - u, upt - w_lo)
+                for x in range(u, u + seg):
+                    covered[x] += 1
+                full = (w_lo == 0) and (seg == upt)
+                if not full:
+                    if w_lo != 0:
+                        assert first, "head partial must be the 1st segment"
+                        assert g not in head_writers
+                        head_writers[g] = (tile, seg)
+                    else:
+                        assert g not in tail_writers
+                        tail_writers[g] = (tile, seg)
+                u += seg
+                first = False
+
+        assert covered == [1] * total, "units must be covered exactly once"
+
+        # fixup reconstruction per tile
+        for t in range(ntiles):
+            g0 = wg_of(t * upt, q, r)
+            gl = wg_of(t * upt + upt - 1, q, r)
+            if g0 == gl:
+                # fully owned: nobody wrote a partial for this tile
+                assert all(tw[0] != t for tw in head_writers.values())
+                assert all(tw[0] != t for tw in tail_writers.values())
+                continue
+            units = tail_writers[g0][1]  # g0's tail partial
+            for gc in range(g0 + 1, gl + 1):
+                assert head_writers[gc][0] == t
+                units += head_writers[gc][1]
+            assert units == upt, f"tile {t}: {units} != {upt}"
