@@ -76,24 +76,25 @@ def choose_tier(m: int, n: int, k: int):
     """Pick the fastest applicable tier for a problem shape, or None when no
     hand-tiled tier divides the shape (callers fall back to rocBLAS).
 
-    Heuristic from the measured 1024..6144 sweep (profiles/overhead_table
-    sweeps): the huge 256x128 macro-tile wins once its grid fills the 512
-    co-resident block slots of the 256-CU chip reasonably evenly; at
-    straggler sizes (e.g. 3072: 288 blocks over 512 slots) the large 64x64
-    tier's fine grid wins; small shapes go to the small/medium tiers."""
+    Thresholds from the final measured 1024..6144 sweep
+    (profiles/cli_sweep_r2.log, stream-K-enabled kernels): the huge
+    256x128 macro-tile wins from ~384 tiles up (its stream-K twin covers
+    the tail-waste sizes), the large 64x64 tier wins the 2048-3072 band
+    (>=1024 blocks), and below that the medium tier's dense grid wins."""
     from .. import kernel_table as kt
 
     def fits(tier):
         t = kt.TILING[tier]
         return m % t["bm"] == 0 and n % t["bn"] == 0 and k % t["bk"] == 0
 
-    if fits("huge"):
-        blocks = (m // 256) * (n // 128)
-        rem = blocks % 512
-        if blocks >= 512 and (rem == 0 or rem >= 256 or blocks >= 2048):
-            return "huge"
-    for tier in ("tall" if m >= 4 * n else "wide" if n >= 4 * m else "large",
-                 "large", "medium", "small"):
+    if fits("huge") and (m // 256) * (n // 128) >= 384:
+        return "huge"
+    skinny = ("tall" if m >= 4 * n else "wide" if n >= 4 * m else None)
+    if skinny and fits(skinny):
+        return skinny
+    if fits("large") and (m // 64) * (n // 64) >= 1024:
+        return "large"
+    for tier in ("medium", "large", "small"):
         if fits(tier):
             return tier
     return None

@@ -65,13 +65,15 @@ def test_choose_tier():
     # big even shapes fill the huge grid
     assert choose_tier(4096, 4096, 4096) == "huge"
     assert choose_tier(8192, 8192, 8192) == "huge"
-    # straggler grid (3072: 288 blocks over 512 slots) -> large
+    assert choose_tier(3584, 3584, 3584) == "huge"
+    # the large tier wins the 2048-3072 band (measured, cli_sweep_r2)
     assert choose_tier(3072, 3072, 3072) == "large"
-    # small/skinny shapes
-    assert choose_tier(1024, 1024, 1024) == "large"
+    assert choose_tier(2048, 2048, 2048) == "large"
+    # below ~1024 large blocks the medium grid wins
+    assert choose_tier(1024, 1024, 1024) == "medium"
     assert choose_tier(512, 64, 256) == "tall"
     assert choose_tier(64, 512, 256) == "wide"
-    assert choose_tier(64, 64, 64) == "large"
+    assert choose_tier(64, 64, 64) == "medium"
     assert choose_tier(32, 32, 32) == "medium"
     assert choose_tier(16, 16, 32) == "small"
     # shapes no tier divides -> None, and the auto entry points fall back
