@@ -32,6 +32,19 @@ print("baseline verdicts (squared residual norms):", res_row, res_col)
 # 5. reference-id dispatch (0=rocBLAS, 1-6 plain, 10 baseline, 11-16 fused)
 ops.run_kernel_id(16, a, b, c)
 
+# 5b. stream-K control: the launcher auto-selects the stream-K twin at
+#     grid-straggler sizes (measured gates); force / disable explicitly:
+import os
+
+os.environ["FT_SGEMM_STREAMK"] = "1"   # force (pipelined callers win from
+ops.sgemm("huge", a, b, c)             # N>=1024 up); "0" disables; unset
+os.environ.pop("FT_SGEMM_STREAMK")     # = auto
+
+# 5c. odd shapes fall back to rocBLAS (FT entry adds the offline ABFT
+#     verdict chain) instead of raising
+ao, bo, co = ops.make_operands(100, 257, 65)
+ops.ft_sgemm_auto(ao, bo, co)
+
 # 6. distributed block-row SGEMM (one process per GPU over RCCL; see
 #    bench.py --mode blockrow for the full multi-rank setup)
 from ft_sgemm_amd.parallel import block_row_sgemm
