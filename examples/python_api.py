@@ -5,6 +5,11 @@ Tensor convention: a column-major MxK fp32 matrix is a contiguous (K, M)
 CUDA tensor (same bytes, zero copies) — see ft_sgemm_amd/__init__.py.
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 from ft_sgemm_amd import ops
@@ -34,8 +39,6 @@ ops.run_kernel_id(16, a, b, c)
 
 # 5b. stream-K control: the launcher auto-selects the stream-K twin at
 #     grid-straggler sizes (measured gates); force / disable explicitly:
-import os
-
 os.environ["FT_SGEMM_STREAMK"] = "1"   # force (pipelined callers win from
 ops.sgemm("huge", a, b, c)             # N>=1024 up); "0" disables; unset
 os.environ.pop("FT_SGEMM_STREAMK")     # = auto
